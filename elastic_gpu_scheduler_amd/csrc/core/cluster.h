@@ -1,0 +1,235 @@
+// Cluster-wide scheduler state.
+//
+// The reference serialises every verb (Assume/Score/Bind/AddPod/ForgetPod)
+// behind one global mutex per scheduler (pkg/scheduler/scheduler.go:44,113)
+// and fans the per-node feasibility check over a fixed 4-goroutine pool
+// (scheduler.go:129-156). Here the node map is read-mostly under a
+// shared_mutex, each node carries its own lock, and the filter fan-out runs
+// on a persistent thread pool sized to the host — concurrent pods scheduling
+// onto different nodes never contend.
+#pragma once
+
+#include <atomic>
+#include <condition_variable>
+#include <functional>
+#include <memory>
+#include <shared_mutex>
+#include <string>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+#include "node.h"
+#include "raters.h"
+#include "types.h"
+
+namespace egs {
+
+// Small fixed thread pool for the filter fan-out.
+class ThreadPool {
+ public:
+  explicit ThreadPool(int n) {
+    if (n <= 0) n = static_cast<int>(std::thread::hardware_concurrency());
+    if (n <= 0) n = 4;
+    for (int i = 0; i < n; ++i)
+      workers_.emplace_back([this] { loop(); });
+  }
+  ~ThreadPool() {
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      stop_ = true;
+    }
+    cv_.notify_all();
+    for (auto& t : workers_) t.join();
+  }
+  int size() const { return static_cast<int>(workers_.size()); }
+
+  // Run fn(i) for i in [0, n) across the pool; blocks until done.
+  void parallel_for(int n, const std::function<void(int)>& fn) {
+    if (n <= 0) return;
+    if (n == 1 || workers_.empty()) {
+      for (int i = 0; i < n; ++i) fn(i);
+      return;
+    }
+    std::atomic<int> next{0};
+    std::atomic<int> done{0};
+    std::mutex done_mu;
+    std::condition_variable done_cv;
+    auto task = [&, n] {
+      int i;
+      while ((i = next.fetch_add(1)) < n) fn(i);
+      if (done.fetch_add(1) + 1 == size()) {
+        std::lock_guard<std::mutex> g(done_mu);
+        done_cv.notify_one();
+      }
+    };
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      for (int i = 0; i < size(); ++i) queue_.push_back(task);
+    }
+    cv_.notify_all();
+    std::unique_lock<std::mutex> g(done_mu);
+    done_cv.wait(g, [&] { return done.load() == size(); });
+  }
+
+ private:
+  void loop() {
+    for (;;) {
+      std::function<void()> task;
+      {
+        std::unique_lock<std::mutex> g(mu_);
+        cv_.wait(g, [this] { return stop_ || !queue_.empty(); });
+        if (stop_ && queue_.empty()) return;
+        task = std::move(queue_.front());
+        queue_.erase(queue_.begin());
+      }
+      task();
+    }
+  }
+  std::vector<std::thread> workers_;
+  std::vector<std::function<void()>> queue_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  bool stop_ = false;
+};
+
+enum class AssumeVerdict : int {
+  kOk = 0,
+  kInfeasible = 1,
+  kUnknownNode = 2,
+};
+
+class ClusterState {
+ public:
+  ClusterState(const std::string& policy, uint64_t seed, int threads)
+      : rater_(make_rater(policy, seed)), pool_(threads) {}
+
+  std::string policy() const { return rater_->name(); }
+
+  void add_node(const std::string& name, std::vector<Device> devices,
+                std::vector<std::vector<int>> topo_hops) {
+    auto alloc = std::make_shared<NodeAllocator>(name, std::move(devices),
+                                                Topology(std::move(topo_hops)));
+    std::unique_lock<std::shared_mutex> g(mu_);
+    nodes_[name] = std::move(alloc);  // replaces any stale entry
+  }
+
+  bool has_node(const std::string& name) {
+    std::shared_lock<std::shared_mutex> g(mu_);
+    return nodes_.count(name) > 0;
+  }
+
+  void remove_node(const std::string& name) {
+    std::unique_lock<std::shared_mutex> g(mu_);
+    nodes_.erase(name);
+  }
+
+  std::vector<std::string> node_names() {
+    std::shared_lock<std::shared_mutex> g(mu_);
+    std::vector<std::string> out;
+    out.reserve(nodes_.size());
+    for (const auto& [n, _] : nodes_) out.push_back(n);
+    return out;
+  }
+
+  // Filter fan-out: feasibility of `req` on every node in `names`.
+  std::vector<int> assume(const std::vector<std::string>& names,
+                          const std::string& uid, const GPURequest& req) {
+    std::vector<std::shared_ptr<NodeAllocator>> allocs(names.size());
+    {
+      std::shared_lock<std::shared_mutex> g(mu_);
+      for (size_t i = 0; i < names.size(); ++i) {
+        auto it = nodes_.find(names[i]);
+        if (it != nodes_.end()) allocs[i] = it->second;
+      }
+    }
+    std::vector<int> verdicts(names.size(), static_cast<int>(AssumeVerdict::kUnknownNode));
+    pool_.parallel_for(static_cast<int>(names.size()), [&](int i) {
+      if (!allocs[i]) return;
+      verdicts[i] = allocs[i]->assume(uid, req, *rater_)
+                        ? static_cast<int>(AssumeVerdict::kOk)
+                        : static_cast<int>(AssumeVerdict::kInfeasible);
+    });
+    return verdicts;
+  }
+
+  std::vector<double> score(const std::vector<std::string>& names,
+                            const std::string& uid, const GPURequest& req) {
+    std::vector<std::shared_ptr<NodeAllocator>> allocs(names.size());
+    {
+      std::shared_lock<std::shared_mutex> g(mu_);
+      for (size_t i = 0; i < names.size(); ++i) {
+        auto it = nodes_.find(names[i]);
+        if (it != nodes_.end()) allocs[i] = it->second;
+      }
+    }
+    std::vector<double> scores(names.size(), kScoreMin);
+    pool_.parallel_for(static_cast<int>(names.size()), [&](int i) {
+      if (!allocs[i]) return;
+      scores[i] = allocs[i]->score(uid, req, *rater_);
+    });
+    return scores;
+  }
+
+  GPUOption allocate(const std::string& node, const std::string& uid,
+                     const GPURequest& req) {
+    auto alloc = get(node);
+    if (!alloc) throw std::runtime_error("unknown node " + node);
+    GPUOption option = alloc->allocate(uid, req, *rater_);
+    {
+      std::lock_guard<std::mutex> g(pod_node_mu_);
+      pod_node_[uid] = node;
+    }
+    return option;
+  }
+
+  void add_pod(const std::string& node, const std::string& uid,
+               const GPURequest& req, const GPUOption& option) {
+    auto alloc = get(node);
+    if (!alloc) throw std::runtime_error("unknown node " + node);
+    alloc->add_pod(uid, req, option);
+    std::lock_guard<std::mutex> g(pod_node_mu_);
+    pod_node_[uid] = node;
+  }
+
+  void note_pod_node(const std::string& uid, const std::string& node) {
+    std::lock_guard<std::mutex> g(pod_node_mu_);
+    pod_node_[uid] = node;
+  }
+
+  void forget_pod(const std::string& uid) {
+    std::string node;
+    {
+      std::lock_guard<std::mutex> g(pod_node_mu_);
+      auto it = pod_node_.find(uid);
+      if (it == pod_node_.end()) return;
+      node = it->second;
+      pod_node_.erase(it);
+    }
+    auto alloc = get(node);
+    if (alloc) alloc->forget_pod(uid);
+  }
+
+  bool known_pod(const std::string& uid) {
+    std::lock_guard<std::mutex> g(pod_node_mu_);
+    return pod_node_.count(uid) > 0;
+  }
+
+  std::shared_ptr<NodeAllocator> get(const std::string& name) {
+    std::shared_lock<std::shared_mutex> g(mu_);
+    auto it = nodes_.find(name);
+    return it == nodes_.end() ? nullptr : it->second;
+  }
+
+  int pool_size() const { return pool_.size(); }
+
+ private:
+  std::unique_ptr<Rater> rater_;
+  ThreadPool pool_;
+  std::shared_mutex mu_;
+  std::unordered_map<std::string, std::shared_ptr<NodeAllocator>> nodes_;
+  std::mutex pod_node_mu_;
+  std::unordered_map<std::string, std::string> pod_node_;  // uid -> node name
+};
+
+}  // namespace egs

@@ -1,0 +1,230 @@
+// Per-node allocator: device accounting + assume cache + commit/cancel.
+//
+// Redesigned from the reference NodeAllocator (pkg/scheduler/node.go:13-160)
+// with deliberate fixes:
+//   * the assume cache is keyed by POD UID, not by a hash of the request
+//     shape — the reference shares one cached option between concurrently
+//     scheduled pods with identical shapes (allocate.go:30-33, node.go:63-72),
+//     which can double-book a card;
+//   * Score never dereferences a missing option (reference nil-deref,
+//     node.go:78-84): a cache miss re-runs the search;
+//   * assumed entries expire after a TTL so pods that are filtered but never
+//     bound don't pin stale placements;
+//   * all verbs are idempotent per pod UID and guarded by a per-node mutex
+//     (the reference serialises the whole cluster behind one global mutex,
+//     scheduler.go:44).
+#pragma once
+
+#include <chrono>
+#include <cstdint>
+#include <mutex>
+#include <optional>
+#include <stdexcept>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "raters.h"
+#include "search.h"
+#include "topology.h"
+#include "types.h"
+
+namespace egs {
+
+constexpr auto kAssumeTTL = std::chrono::seconds(300);
+
+class NodeAllocator {
+ public:
+  NodeAllocator(std::string name, std::vector<Device> devices, Topology topo)
+      : name_(std::move(name)), devices_(std::move(devices)), topo_(std::move(topo)) {}
+
+  const std::string& name() const { return name_; }
+  int num_devices() const { return static_cast<int>(devices_.size()); }
+
+  // Feasibility + placement, cached per pod UID. Returns true if a placement
+  // exists (and remembers it for the later Score/Allocate of the same pod).
+  bool assume(const std::string& uid, const GPURequest& req, const Rater& rater) {
+    std::lock_guard<std::mutex> g(mu_);
+    gc_assumed_locked();
+    auto it = assumed_.find(uid);
+    if (it != assumed_.end()) return true;
+    auto res = run_search_locked(req, rater);
+    if (!res.feasible) return false;
+    assumed_[uid] = {std::move(res.option), now()};
+    return true;
+  }
+
+  // Score for prioritize. A cache miss re-runs the search (never crashes on a
+  // missing option). Returns kScoreMin when infeasible.
+  double score(const std::string& uid, const GPURequest& req, const Rater& rater) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = assumed_.find(uid);
+    if (it != assumed_.end()) return it->second.option.score;
+    auto res = run_search_locked(req, rater);
+    if (!res.feasible) return kScoreMin;
+    assumed_[uid] = {res.option, now()};
+    return res.option.score;
+  }
+
+  // Commit the assumed placement for a pod (bind path). If no assumed entry
+  // exists (e.g. scheduler restarted between filter and bind), a fresh search
+  // runs. Throws std::runtime_error when infeasible.
+  GPUOption allocate(const std::string& uid, const GPURequest& req, const Rater& rater) {
+    std::lock_guard<std::mutex> g(mu_);
+    if (pods_.count(uid)) return pods_[uid];  // idempotent re-bind
+    GPUOption option;
+    auto it = assumed_.find(uid);
+    if (it != assumed_.end()) {
+      option = it->second.option;
+      assumed_.erase(it);
+      // Re-validate: the world may have changed since Assume.
+      if (!fits_locked(req, option)) {
+        auto res = run_search_locked(req, rater);
+        if (!res.feasible) throw std::runtime_error("insufficient GPU resources on " + name_);
+        option = std::move(res.option);
+      }
+    } else {
+      auto res = run_search_locked(req, rater);
+      if (!res.feasible) throw std::runtime_error("insufficient GPU resources on " + name_);
+      option = std::move(res.option);
+    }
+    transact_locked(req, option);
+    pods_[uid] = option;
+    requests_[uid] = req;
+    return option;
+  }
+
+  // Replay a known placement (controller sync / crash recovery). Idempotent.
+  // Throws if the placement no longer fits (double-booked annotations).
+  void add_pod(const std::string& uid, const GPURequest& req, const GPUOption& option) {
+    std::lock_guard<std::mutex> g(mu_);
+    if (pods_.count(uid)) return;
+    if (!fits_locked(req, option))
+      throw std::runtime_error("placement replay does not fit on " + name_);
+    transact_locked(req, option);
+    pods_[uid] = option;
+    requests_[uid] = req;
+  }
+
+  // Release a pod's resources. Idempotent; unknown UIDs are a no-op.
+  void forget_pod(const std::string& uid) {
+    std::lock_guard<std::mutex> g(mu_);
+    assumed_.erase(uid);
+    auto it = pods_.find(uid);
+    if (it == pods_.end()) return;
+    cancel_locked(requests_[uid], it->second);
+    pods_.erase(it);
+    requests_.erase(uid);
+  }
+
+  bool known_pod(const std::string& uid) {
+    std::lock_guard<std::mutex> g(mu_);
+    return pods_.count(uid) > 0;
+  }
+
+  std::vector<Device> snapshot() {
+    std::lock_guard<std::mutex> g(mu_);
+    return devices_;
+  }
+
+  std::vector<std::string> pod_uids() {
+    std::lock_guard<std::mutex> g(mu_);
+    std::vector<std::string> out;
+    out.reserve(pods_.size());
+    for (const auto& [uid, _] : pods_) out.push_back(uid);
+    return out;
+  }
+
+  const Topology& topology() const { return topo_; }
+
+ private:
+  using Clock = std::chrono::steady_clock;
+  struct Assumed {
+    GPUOption option;
+    Clock::time_point at;
+  };
+
+  static Clock::time_point now() { return Clock::now(); }
+
+  SearchResult run_search_locked(const GPURequest& req, const Rater& rater) {
+    RateContext ctx;
+    ctx.devices = &devices_;
+    ctx.topo = &topo_;
+    ctx.salt = detail::fnv1a(1469598103ULL, std::hash<std::string>{}(name_));
+    return search_placement(devices_, req, rater, ctx);
+  }
+
+  bool fits_locked(const GPURequest& req, const GPUOption& option) const {
+    std::vector<Device> copy = devices_;
+    for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
+      const GPUUnit& u = req[c];
+      for (int idx : option.allocated[c]) {
+        if (idx < 0 || idx >= static_cast<int>(copy.size())) return false;
+        Device& d = copy[idx];
+        if (u.whole_cards()) {
+          if (!d.whole_free()) return false;
+          d.core_avail = 0;
+          d.mem_avail = 0;
+        } else {
+          if (!d.can_fit(u.core, u.memory)) return false;
+          d.core_avail -= u.core;
+          d.mem_avail -= u.memory;
+        }
+      }
+    }
+    return true;
+  }
+
+  void transact_locked(const GPURequest& req, const GPUOption& option) {
+    for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
+      const GPUUnit& u = req[c];
+      for (int idx : option.allocated[c]) {
+        Device& d = devices_[idx];
+        if (u.whole_cards()) {
+          d.core_avail = 0;
+          d.mem_avail = 0;
+        } else {
+          d.core_avail -= u.core;
+          d.mem_avail -= u.memory;
+        }
+      }
+    }
+  }
+
+  void cancel_locked(const GPURequest& req, const GPUOption& option) {
+    for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
+      const GPUUnit& u = req[c];
+      for (int idx : option.allocated[c]) {
+        Device& d = devices_[idx];
+        if (u.whole_cards()) {
+          d.core_avail = d.core_total;
+          d.mem_avail = d.mem_total;
+        } else {
+          d.core_avail = std::min(d.core_total, d.core_avail + u.core);
+          d.mem_avail = std::min(d.mem_total, d.mem_avail + u.memory);
+        }
+      }
+    }
+  }
+
+  void gc_assumed_locked() {
+    if (assumed_.size() < 1024) return;  // amortise: only sweep when large
+    auto cutoff = now() - kAssumeTTL;
+    for (auto it = assumed_.begin(); it != assumed_.end();) {
+      if (it->second.at < cutoff)
+        it = assumed_.erase(it);
+      else
+        ++it;
+    }
+  }
+
+  std::string name_;
+  std::vector<Device> devices_;
+  Topology topo_;
+  std::mutex mu_;
+  std::unordered_map<std::string, Assumed> assumed_;      // uid -> pending placement
+  std::unordered_map<std::string, GPUOption> pods_;       // uid -> committed placement
+  std::unordered_map<std::string, GPURequest> requests_;  // uid -> demand (for cancel)
+};
+
+}  // namespace egs

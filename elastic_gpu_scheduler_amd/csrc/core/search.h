@@ -1,0 +1,193 @@
+// Placement search: choose device indexes for every container of a pod.
+//
+// Semantics follow the reference's GPUs.Trade (pkg/scheduler/gpu.go:65-129):
+//   * a whole-card container (gpu_count > 0) takes N fully-free cards
+//     exclusively;
+//   * a fractional container takes a share (core%, memory bytes) of exactly
+//     one card; multiple containers may share a card;
+//   * every complete assignment is scored by the Rater and the best is kept.
+// Improvements over the reference:
+//   * whole-card sets are CHOSEN (xGMI-topology- and policy-scored k-subsets),
+//     not just the first N free cards (gpu.go:96-108);
+//   * symmetric fractional branches are deduplicated and the DFS is bounded
+//     by a deterministic leaf budget with greedy candidate ordering, so worst
+//     cases stay microseconds instead of cards^containers;
+//   * ties break deterministically (first-best in candidate order), so
+//     Assume/Score/Bind always agree on the same placement.
+#pragma once
+
+#include <algorithm>
+#include <cstdint>
+#include <functional>
+#include <utility>
+#include <vector>
+
+#include "raters.h"
+#include "topology.h"
+#include "types.h"
+
+namespace egs {
+
+// Upper bound on scored complete assignments per search. Candidates are
+// ordered greedily, so truncation degrades score quality, never feasibility
+// (the first feasible leaf is always reached if one exists in the explored
+// prefix; the greedy order explores plausible placements first).
+constexpr int kMaxLeafEvals = 4096;
+// Upper bound on whole-card k-subsets enumerated per container.
+constexpr int kMaxWholeCardCandidates = 128;
+
+struct SearchResult {
+  bool feasible = false;
+  GPUOption option;
+  int leaves_evaluated = 0;
+};
+
+namespace search_detail {
+
+// Enumerate k-subsets of `free_cards`, best-topology-first, capped.
+inline void enumerate_subsets(const std::vector<int>& free_cards, int k,
+                              const Topology& topo,
+                              std::vector<std::vector<int>>* out) {
+  const int n = static_cast<int>(free_cards.size());
+  if (k > n) return;
+  std::vector<int> pick;
+  pick.reserve(k);
+  struct Cand {
+    std::vector<int> cards;
+    int cost;
+  };
+  std::vector<Cand> cands;
+  // DFS over combinations (lexicographic, deterministic).
+  std::function<void(int)> rec = [&](int start) {
+    if (static_cast<int>(pick.size()) == k) {
+      std::vector<int> cards;
+      cards.reserve(k);
+      for (int i : pick) cards.push_back(free_cards[i]);
+      int cost = topo.set_cost(cards);
+      cands.push_back({std::move(cards), cost});
+      return;
+    }
+    if (static_cast<int>(cands.size()) >= kMaxWholeCardCandidates * 8) return;
+    for (int i = start; i < n; ++i) {
+      pick.push_back(i);
+      rec(i + 1);
+      pick.pop_back();
+    }
+  };
+  rec(0);
+  std::stable_sort(cands.begin(), cands.end(),
+                   [](const Cand& a, const Cand& b) { return a.cost < b.cost; });
+  int keep = std::min<int>(static_cast<int>(cands.size()), kMaxWholeCardCandidates);
+  for (int i = 0; i < keep; ++i) out->push_back(std::move(cands[i].cards));
+}
+
+struct DfsState {
+  std::vector<Device> devices;
+  const GPURequest* req;
+  const Rater* rater;
+  const RateContext* ctx;
+  GPUOption current;
+  GPUOption best;
+  bool found = false;
+  int leaves = 0;
+};
+
+inline void dfs(DfsState& st, size_t c);
+
+inline void try_candidate(DfsState& st, size_t c, const std::vector<int>& cards,
+                          bool whole) {
+  const GPUUnit& u = (*st.req)[c];
+  // Apply.
+  std::vector<std::pair<int, Device>> saved;
+  saved.reserve(cards.size());
+  for (int idx : cards) {
+    saved.emplace_back(idx, st.devices[idx]);
+    Device& d = st.devices[idx];
+    if (whole) {
+      d.core_avail = 0;
+      d.mem_avail = 0;
+    } else {
+      d.core_avail -= u.core;
+      d.mem_avail -= u.memory;
+    }
+  }
+  st.current.allocated[c] = cards;
+  dfs(st, c + 1);
+  // Undo.
+  st.current.allocated[c].clear();
+  for (auto& [idx, d] : saved) st.devices[idx] = d;
+}
+
+inline void dfs(DfsState& st, size_t c) {
+  if (st.leaves >= kMaxLeafEvals) return;
+  const GPURequest& req = *st.req;
+  if (c == req.size()) {
+    ++st.leaves;
+    double score = st.rater->rate(*st.ctx, req, st.current);
+    // Strictly-greater keeps the FIRST best in deterministic candidate order.
+    if (!st.found || score > st.best.score) {
+      st.best = st.current;
+      st.best.score = score;
+      st.found = true;
+    }
+    return;
+  }
+  const GPUUnit& u = req[c];
+  if (!u.needs_gpu()) {
+    st.current.allocated[c].clear();
+    dfs(st, c + 1);
+    return;
+  }
+  if (u.whole_cards()) {
+    std::vector<int> free_cards;
+    for (int i = 0; i < static_cast<int>(st.devices.size()); ++i)
+      if (st.devices[i].whole_free()) free_cards.push_back(i);
+    if (static_cast<int>(free_cards.size()) < u.gpu_count) return;  // infeasible here
+    std::vector<std::vector<int>> subsets;
+    const Topology* topo = st.ctx->topo;
+    static const Topology kEmpty;
+    enumerate_subsets(free_cards, u.gpu_count, topo ? *topo : kEmpty, &subsets);
+    for (const auto& cards : subsets) {
+      try_candidate(st, c, cards, /*whole=*/true);
+      if (st.leaves >= kMaxLeafEvals) return;
+    }
+  } else {
+    // Fractional: one card among those that fit; dedupe cards with an
+    // identical availability signature when topology is uniform (symmetric
+    // branches produce identical scores).
+    bool uniform = st.ctx->topo == nullptr || st.ctx->topo->empty();
+    std::vector<std::pair<int64_t, int64_t>> seen;
+    for (int i = 0; i < static_cast<int>(st.devices.size()); ++i) {
+      const Device& d = st.devices[i];
+      if (!d.can_fit(u.core, u.memory)) continue;
+      if (uniform) {
+        auto sig = std::make_pair(static_cast<int64_t>(d.core_avail), d.mem_avail);
+        if (std::find(seen.begin(), seen.end(), sig) != seen.end()) continue;
+        seen.push_back(sig);
+      }
+      try_candidate(st, c, {i}, /*whole=*/false);
+      if (st.leaves >= kMaxLeafEvals) return;
+    }
+  }
+}
+
+}  // namespace search_detail
+
+inline SearchResult search_placement(const std::vector<Device>& devices,
+                                     const GPURequest& req, const Rater& rater,
+                                     const RateContext& ctx) {
+  search_detail::DfsState st;
+  st.devices = devices;
+  st.req = &req;
+  st.rater = &rater;
+  st.ctx = &ctx;
+  st.current.allocated.resize(req.size());
+  search_detail::dfs(st, 0);
+  SearchResult res;
+  res.feasible = st.found;
+  res.option = std::move(st.best);
+  res.leaves_evaluated = st.leaves;
+  return res;
+}
+
+}  // namespace egs
