@@ -1,0 +1,139 @@
+"""Reconcile controller: keeps scheduler accounting honest against the
+apiserver.
+
+Analogue of the reference's informer controller (pkg/controller/
+controller.go): watches pods, filters to GPU pods, runs a rate-limited
+workqueue with N workers, and reconciles lifecycle into the scheduler —
+AddPod for running assigned pods, ForgetPod for completed/deleted ones
+(controller.go:154-185, 301-331). Differences:
+
+  * a plain threaded workqueue with per-key dedup instead of client-go
+    machinery; resync is an explicit periodic relist (reference uses a 30 s
+    shared-informer resync, controller.go:24);
+  * the node informer the reference creates but never consults
+    (controller.go:97-99) is replaced by node-cache invalidation: node
+    deletions/changes evict the per-node allocator so inventory updates
+    (e.g. the agent republishing topology) take effect.
+"""
+from __future__ import annotations
+
+import logging
+import queue
+import threading
+import time
+from typing import Any, Dict, Optional
+
+from elastic_gpu_scheduler_amd.k8s import objects as obj
+from elastic_gpu_scheduler_amd.k8s.client import KubeClient, NotFoundError
+from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+
+log = logging.getLogger("egs.controller")
+
+
+class Controller:
+    def __init__(self, client: KubeClient, registry: SchedulerRegistry,
+                 workers: int = 1, resync_seconds: float = 30.0) -> None:
+        self.client = client
+        self.registry = registry
+        self.workers = max(1, workers)
+        self.resync_seconds = resync_seconds
+        self._queue: "queue.Queue[Optional[str]]" = queue.Queue()
+        self._pending: Dict[str, Dict[str, Any]] = {}  # key -> last seen pod
+        self._pending_mu = threading.Lock()
+        self._stop = threading.Event()
+        self._threads: list[threading.Thread] = []
+        self._unsubscribe = None
+
+    # -- lifecycle --
+
+    def start(self) -> None:
+        self._unsubscribe = self.client.watch_pods(self._on_event)
+        for i in range(self.workers):
+            t = threading.Thread(target=self._worker, name=f"egs-sync-{i}",
+                                 daemon=True)
+            t.start()
+            self._threads.append(t)
+        t = threading.Thread(target=self._resync_loop, name="egs-resync",
+                             daemon=True)
+        t.start()
+        self._threads.append(t)
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._unsubscribe:
+            self._unsubscribe()
+        for _ in self._threads:
+            self._queue.put(None)
+        for t in self._threads:
+            t.join(timeout=2.0)
+
+    # -- watch handling (reference controller.go:212-299) --
+
+    def _on_event(self, event_type: str, pod: Dict[str, Any]) -> None:
+        if not obj.is_gpu_pod(pod):
+            return
+        key = obj.pod_key(pod)
+        if event_type == "DELETED":
+            pod = dict(pod)
+            pod.setdefault("metadata", {})["_egs_deleted"] = True
+        self._enqueue(key, pod)
+
+    def _enqueue(self, key: str, pod: Dict[str, Any]) -> None:
+        with self._pending_mu:
+            fresh = key not in self._pending
+            self._pending[key] = pod
+        if fresh:
+            self._queue.put(key)
+
+    # -- workers (reference processNextWorkItem, controller.go:189-210) --
+
+    def _worker(self) -> None:
+        while not self._stop.is_set():
+            key = self._queue.get()
+            if key is None:
+                return
+            with self._pending_mu:
+                pod = self._pending.pop(key, None)
+            if pod is None:
+                continue
+            try:
+                self._sync_pod(pod)
+            except Exception:
+                log.exception("sync of %s failed", key)
+
+    def _sync_pod(self, pod: Dict[str, Any]) -> None:
+        """Reference syncPod (controller.go:154-185): completed/deleted ->
+        release; running & assigned -> account."""
+        sch = self.registry.for_pod(pod)
+        if sch is None:
+            return
+        deleted = pod.get("metadata", {}).get("_egs_deleted", False)
+        if deleted or obj.is_completed_pod(pod):
+            sch.forget_pod(pod)
+            return
+        if obj.pod_node_name(pod) and obj.is_assumed(pod):
+            sch.add_pod(pod)
+
+    # -- periodic resync (stand-in for informer resync) --
+
+    def _resync_loop(self) -> None:
+        while not self._stop.wait(self.resync_seconds):
+            try:
+                self.resync_once()
+            except Exception:
+                log.exception("resync failed")
+
+    def resync_once(self) -> None:
+        pods = self.client.list_pods()
+        live = set()
+        for pod in pods:
+            if not obj.is_gpu_pod(pod):
+                continue
+            live.add(obj.pod_uid(pod))
+            self._sync_pod(pod)
+        # Evict accounting for pods that vanished without a DELETE event.
+        for sch in {id(s): s for s in self.registry.schedulers.values()}.values():
+            for node in sch.state.node_names():
+                for uid in sch.state.node_pods(node):
+                    if uid not in live:
+                        sch.state.forget_pod(uid)

@@ -1,0 +1,366 @@
+"""Kubernetes clients: the interface, an in-memory fake, and a real REST client.
+
+The reference takes *kubernetes.Clientset concretely (pkg/scheduler/
+scheduler.go:24), which blocks fake injection — SURVEY.md flags that as a
+design flaw. Here everything programs against KubeClient, and FakeKubeClient
+implements the full optimistic-concurrency + watch semantics the scheduler
+and controller rely on, so the whole control plane is testable offline.
+"""
+from __future__ import annotations
+
+import copy
+import json
+import threading
+import uuid
+from typing import Any, Callable, Dict, List, Optional
+
+Pod = Dict[str, Any]
+Node = Dict[str, Any]
+WatchHandler = Callable[[str, Pod], None]  # (event_type, object)
+
+
+class ConflictError(Exception):
+    """Optimistic-lock failure (HTTP 409). The reference matches the error
+    *text* (pkg/utils/types.go:15); we use a typed error."""
+
+
+class NotFoundError(Exception):
+    """HTTP 404."""
+
+
+class KubeClient:
+    """Minimal apiserver surface the scheduler/controller/agent need."""
+
+    # pods
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        raise NotImplementedError
+
+    def list_pods(self, label_selector: Optional[Dict[str, str]] = None,
+                  field_selector: Optional[Dict[str, str]] = None) -> List[Pod]:
+        raise NotImplementedError
+
+    def update_pod(self, pod: Pod) -> Pod:
+        raise NotImplementedError
+
+    def bind_pod(self, namespace: str, name: str, node: str) -> None:
+        raise NotImplementedError
+
+    # nodes
+    def get_node(self, name: str) -> Node:
+        raise NotImplementedError
+
+    def list_nodes(self) -> List[Node]:
+        raise NotImplementedError
+
+    def patch_node_annotations(self, name: str, annotations: Dict[str, str]) -> Node:
+        raise NotImplementedError
+
+    # events (the reference wires a recorder but never emits — controller.go:57-65;
+    # we actually emit scheduling events)
+    def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
+        raise NotImplementedError
+
+    # watch
+    def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
+        """Register a pod watch; returns an unsubscribe callable."""
+        raise NotImplementedError
+
+
+def _match_labels(obj: Dict[str, Any], selector: Dict[str, str]) -> bool:
+    labels = obj.get("metadata", {}).get("labels", {}) or {}
+    return all(labels.get(k) == v for k, v in selector.items())
+
+
+class FakeKubeClient(KubeClient):
+    """In-memory apiserver with resourceVersion optimistic locking and watch
+    fan-out. Mirrors the semantics of k8s.io/client-go/kubernetes/fake closely
+    enough for scheduler/controller tests and the bench harness."""
+
+    def __init__(self) -> None:
+        self._mu = threading.RLock()
+        self._pods: Dict[str, Pod] = {}   # "ns/name" -> pod
+        self._nodes: Dict[str, Node] = {}
+        self._events: List[Dict[str, Any]] = []
+        self._rv = 0
+        self._watchers: List[WatchHandler] = []
+
+    # -- helpers --
+    def _next_rv(self) -> str:
+        self._rv += 1
+        return str(self._rv)
+
+    def _notify(self, event_type: str, pod: Pod) -> None:
+        for h in list(self._watchers):
+            h(event_type, copy.deepcopy(pod))
+
+    @staticmethod
+    def _key(namespace: str, name: str) -> str:
+        return f"{namespace}/{name}"
+
+    # -- seeding (test/bench setup) --
+    def add_node(self, node: Node) -> Node:
+        with self._mu:
+            node = copy.deepcopy(node)
+            meta = node.setdefault("metadata", {})
+            meta.setdefault("uid", str(uuid.uuid4()))
+            meta["resourceVersion"] = self._next_rv()
+            self._nodes[meta["name"]] = node
+            return copy.deepcopy(node)
+
+    def create_pod(self, pod: Pod) -> Pod:
+        with self._mu:
+            pod = copy.deepcopy(pod)
+            meta = pod.setdefault("metadata", {})
+            meta.setdefault("namespace", "default")
+            meta.setdefault("uid", str(uuid.uuid4()))
+            meta["resourceVersion"] = self._next_rv()
+            key = self._key(meta["namespace"], meta["name"])
+            if key in self._pods:
+                raise ConflictError(f"pod {key} already exists")
+            self._pods[key] = pod
+            out = copy.deepcopy(pod)
+        self._notify("ADDED", out)
+        return out
+
+    def delete_pod(self, namespace: str, name: str) -> None:
+        with self._mu:
+            key = self._key(namespace, name)
+            pod = self._pods.pop(key, None)
+        if pod is not None:
+            self._notify("DELETED", copy.deepcopy(pod))
+
+    def set_pod_phase(self, namespace: str, name: str, phase: str) -> None:
+        with self._mu:
+            key = self._key(namespace, name)
+            if key not in self._pods:
+                raise NotFoundError(key)
+            self._pods[key].setdefault("status", {})["phase"] = phase
+            self._pods[key]["metadata"]["resourceVersion"] = self._next_rv()
+            pod = copy.deepcopy(self._pods[key])
+        self._notify("MODIFIED", pod)
+
+    # -- KubeClient impl --
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        with self._mu:
+            key = self._key(namespace, name)
+            if key not in self._pods:
+                raise NotFoundError(f"pod {key} not found")
+            return copy.deepcopy(self._pods[key])
+
+    def list_pods(self, label_selector: Optional[Dict[str, str]] = None,
+                  field_selector: Optional[Dict[str, str]] = None) -> List[Pod]:
+        with self._mu:
+            pods = [copy.deepcopy(p) for p in self._pods.values()]
+        if label_selector:
+            pods = [p for p in pods if _match_labels(p, label_selector)]
+        if field_selector:
+            node = field_selector.get("spec.nodeName")
+            if node is not None:
+                pods = [p for p in pods
+                        if p.get("spec", {}).get("nodeName") == node]
+        return pods
+
+    def update_pod(self, pod: Pod) -> Pod:
+        with self._mu:
+            meta = pod.get("metadata", {})
+            key = self._key(meta.get("namespace", "default"), meta.get("name", ""))
+            if key not in self._pods:
+                raise NotFoundError(f"pod {key} not found")
+            current = self._pods[key]
+            if meta.get("resourceVersion") != current["metadata"]["resourceVersion"]:
+                raise ConflictError(
+                    f"pod {key}: resourceVersion mismatch "
+                    f"({meta.get('resourceVersion')} != "
+                    f"{current['metadata']['resourceVersion']})")
+            pod = copy.deepcopy(pod)
+            pod["metadata"]["resourceVersion"] = self._next_rv()
+            self._pods[key] = pod
+            out = copy.deepcopy(pod)
+        self._notify("MODIFIED", out)
+        return out
+
+    def bind_pod(self, namespace: str, name: str, node: str) -> None:
+        with self._mu:
+            key = self._key(namespace, name)
+            if key not in self._pods:
+                raise NotFoundError(f"pod {key} not found")
+            if node not in self._nodes:
+                raise NotFoundError(f"node {node} not found")
+            self._pods[key].setdefault("spec", {})["nodeName"] = node
+            self._pods[key]["metadata"]["resourceVersion"] = self._next_rv()
+            pod = copy.deepcopy(self._pods[key])
+        self._notify("MODIFIED", pod)
+
+    def get_node(self, name: str) -> Node:
+        with self._mu:
+            if name not in self._nodes:
+                raise NotFoundError(f"node {name} not found")
+            return copy.deepcopy(self._nodes[name])
+
+    def list_nodes(self) -> List[Node]:
+        with self._mu:
+            return [copy.deepcopy(n) for n in self._nodes.values()]
+
+    def patch_node_annotations(self, name: str, annotations: Dict[str, str]) -> Node:
+        with self._mu:
+            if name not in self._nodes:
+                raise NotFoundError(f"node {name} not found")
+            node = self._nodes[name]
+            node.setdefault("metadata", {}).setdefault("annotations", {}).update(
+                annotations)
+            node["metadata"]["resourceVersion"] = self._next_rv()
+            return copy.deepcopy(node)
+
+    def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
+        with self._mu:
+            self._events.append(copy.deepcopy(event))
+
+    @property
+    def events(self) -> List[Dict[str, Any]]:
+        with self._mu:
+            return list(self._events)
+
+    def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
+        with self._mu:
+            self._watchers.append(handler)
+
+        def unsubscribe() -> None:
+            with self._mu:
+                if handler in self._watchers:
+                    self._watchers.remove(handler)
+
+        return unsubscribe
+
+
+class RealKubeClient(KubeClient):
+    """REST client for a live apiserver (in-cluster or kubeconfig).
+
+    Analogue of the reference's client-go bootstrap (pkg/utils/utils.go:44-68).
+    Uses httpx synchronously; watch_pods runs a background streaming watch
+    with automatic reconnect. This class is exercised against HTTP mocks in
+    tests (no live cluster in CI).
+    """
+
+    def __init__(self, base_url: str, token: Optional[str] = None,
+                 verify: "bool | str" = True) -> None:
+        import httpx
+
+        headers = {"Content-Type": "application/json"}
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        self._client = httpx.Client(base_url=base_url, headers=headers,
+                                    verify=verify, timeout=30.0)
+        self._watch_stop = threading.Event()
+
+    @classmethod
+    def from_env(cls) -> "RealKubeClient":
+        """In-cluster service account, else $KUBECONFIG / ~/.kube/config."""
+        import os
+        from pathlib import Path
+
+        sa = Path("/var/run/secrets/kubernetes.io/serviceaccount")
+        if (sa / "token").exists():
+            host = os.environ.get("KUBERNETES_SERVICE_HOST", "kubernetes.default.svc")
+            port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+            return cls(f"https://{host}:{port}",
+                       token=(sa / "token").read_text().strip(),
+                       verify=str(sa / "ca.crt"))
+        cfg_path = os.environ.get("KUBECONFIG", str(Path.home() / ".kube" / "config"))
+        import yaml
+
+        cfg = yaml.safe_load(Path(cfg_path).read_text())
+        ctx_name = cfg.get("current-context")
+        ctx = next(c["context"] for c in cfg["contexts"] if c["name"] == ctx_name)
+        cluster = next(c["cluster"] for c in cfg["clusters"]
+                       if c["name"] == ctx["cluster"])
+        user = next(u["user"] for u in cfg["users"] if u["name"] == ctx["user"])
+        token = user.get("token")
+        return cls(cluster["server"], token=token,
+                   verify=cluster.get("certificate-authority", True))
+
+    def _check(self, resp) -> Any:
+        if resp.status_code == 404:
+            raise NotFoundError(resp.text)
+        if resp.status_code == 409:
+            raise ConflictError(resp.text)
+        resp.raise_for_status()
+        return resp.json() if resp.content else None
+
+    def get_pod(self, namespace: str, name: str) -> Pod:
+        return self._check(self._client.get(
+            f"/api/v1/namespaces/{namespace}/pods/{name}"))
+
+    def list_pods(self, label_selector: Optional[Dict[str, str]] = None,
+                  field_selector: Optional[Dict[str, str]] = None) -> List[Pod]:
+        params = {}
+        if label_selector:
+            params["labelSelector"] = ",".join(f"{k}={v}"
+                                               for k, v in label_selector.items())
+        if field_selector:
+            params["fieldSelector"] = ",".join(f"{k}={v}"
+                                               for k, v in field_selector.items())
+        out = self._check(self._client.get("/api/v1/pods", params=params))
+        return out.get("items", [])
+
+    def update_pod(self, pod: Pod) -> Pod:
+        meta = pod["metadata"]
+        return self._check(self._client.put(
+            f"/api/v1/namespaces/{meta.get('namespace', 'default')}/pods/"
+            f"{meta['name']}", content=json.dumps(pod)))
+
+    def bind_pod(self, namespace: str, name: str, node: str) -> None:
+        binding = {
+            "apiVersion": "v1",
+            "kind": "Binding",
+            "metadata": {"name": name, "namespace": namespace},
+            "target": {"apiVersion": "v1", "kind": "Node", "name": node},
+        }
+        self._check(self._client.post(
+            f"/api/v1/namespaces/{namespace}/pods/{name}/binding",
+            content=json.dumps(binding)))
+
+    def get_node(self, name: str) -> Node:
+        return self._check(self._client.get(f"/api/v1/nodes/{name}"))
+
+    def list_nodes(self) -> List[Node]:
+        out = self._check(self._client.get("/api/v1/nodes"))
+        return out.get("items", [])
+
+    def patch_node_annotations(self, name: str, annotations: Dict[str, str]) -> Node:
+        patch = {"metadata": {"annotations": annotations}}
+        return self._check(self._client.patch(
+            f"/api/v1/nodes/{name}", content=json.dumps(patch),
+            headers={"Content-Type": "application/strategic-merge-patch+json"}))
+
+    def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
+        self._check(self._client.post(
+            f"/api/v1/namespaces/{namespace}/events", content=json.dumps(event)))
+
+    def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
+        stop = threading.Event()
+
+        def run() -> None:
+            while not stop.is_set():
+                try:
+                    with self._client.stream(
+                            "GET", "/api/v1/pods",
+                            params={"watch": "true"}, timeout=None) as resp:
+                        for line in resp.iter_lines():
+                            if stop.is_set():
+                                return
+                            if not line:
+                                continue
+                            evt = json.loads(line)
+                            handler(evt.get("type", ""), evt.get("object", {}))
+                except Exception:
+                    if stop.is_set():
+                        return
+                    stop.wait(1.0)  # reconnect backoff
+
+        thread = threading.Thread(target=run, name="egs-watch", daemon=True)
+        thread.start()
+
+        def unsubscribe() -> None:
+            stop.set()
+
+        return unsubscribe

@@ -1,0 +1,197 @@
+"""ASGI application implementing the kube-scheduler extender protocol.
+
+Wire compatibility: k8s.io/kube-scheduler/extender/v1 JSON — the same routes
+and payloads the reference serves (pkg/routes/routes.go):
+
+  POST /scheduler/filter      ExtenderArgs        -> ExtenderFilterResult
+  POST /scheduler/priorities  ExtenderArgs        -> HostPriorityList
+  POST /scheduler/bind        ExtenderBindingArgs -> ExtenderBindingResult
+  GET  /scheduler/status      per-node accounting JSON
+  GET  /version
+  GET  /metrics               Prometheus text (new)
+  GET  /debug/stacks          all-thread dump (analogue of /debug/pprof)
+  GET  /healthz
+
+Fixes over the reference: malformed JSON returns 400 instead of panicking
+(routes.go:97-103); filter without NodeNames returns a structured error (we
+require nodeCacheCapable=true exactly like routes.go:59-64).
+
+Implemented as a plain ASGI callable (no framework) so the hot path is one
+json.loads + the native core call + one json.dumps; serve with uvicorn.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import sys
+import traceback
+from typing import Any, Dict
+
+from elastic_gpu_scheduler_amd.k8s import objects as obj
+from elastic_gpu_scheduler_amd.k8s.client import NotFoundError
+from elastic_gpu_scheduler_amd.scheduler.service import BindError, SchedulerRegistry
+from elastic_gpu_scheduler_amd.utils import metrics
+from elastic_gpu_scheduler_amd.version import __version__
+
+log = logging.getLogger("egs.server")
+
+
+class ExtenderApp:
+    def __init__(self, registry: SchedulerRegistry) -> None:
+        self.registry = registry
+        self._routes = {
+            ("POST", "/scheduler/filter"): self.filter,
+            ("POST", "/scheduler/priorities"): self.priorities,
+            ("POST", "/scheduler/bind"): self.bind,
+            ("GET", "/scheduler/status"): self.status,
+            ("GET", "/version"): self.version,
+            ("GET", "/metrics"): self.metrics,
+            ("GET", "/healthz"): self.healthz,
+            ("GET", "/debug/stacks"): self.debug_stacks,
+        }
+
+    # ---- ASGI plumbing ---------------------------------------------------
+
+    async def __call__(self, scope, receive, send) -> None:
+        if scope["type"] != "http":
+            return
+        method = scope["method"]
+        path = scope["path"]
+        handler = self._routes.get((method, path))
+        if handler is None:
+            await self._respond(send, 404, {"error": f"no route {method} {path}"})
+            return
+        body = b""
+        while True:
+            msg = await receive()
+            if msg["type"] == "http.request":
+                body += msg.get("body", b"")
+                if not msg.get("more_body"):
+                    break
+            else:
+                break
+        verb = path.rsplit("/", 1)[-1]
+        with metrics.VERB_LATENCY.labels(verb).time():
+            try:
+                status, payload, raw = handler(body)
+                metrics.REQUESTS.labels(verb, "ok" if status < 400 else "error").inc()
+            except _BadRequest as exc:
+                status, payload, raw = 400, {"error": str(exc)}, None
+                metrics.REQUESTS.labels(verb, "bad_request").inc()
+            except Exception as exc:  # never crash the server on one request
+                log.exception("%s %s failed", method, path)
+                status, payload, raw = 500, {"error": f"{type(exc).__name__}: {exc}"}, None
+                metrics.REQUESTS.labels(verb, "exception").inc()
+        await self._respond(send, status, payload, raw)
+
+    async def _respond(self, send, status: int, payload=None, raw: bytes = None,
+                       content_type: bytes = b"application/json") -> None:
+        body = raw if raw is not None else json.dumps(payload).encode()
+        await send({"type": "http.response.start", "status": status,
+                    "headers": [(b"content-type", content_type)]})
+        await send({"type": "http.response.body", "body": body})
+
+    # ---- handlers --------------------------------------------------------
+
+    def filter(self, body: bytes):
+        args = _parse_json(body)
+        pod = args.get("pod")
+        if not pod:
+            raise _BadRequest("ExtenderArgs.pod missing")
+        node_names = args.get("nodenames")
+        if node_names is None:
+            # Reference behavior (routes.go:59-64): require nodeCacheCapable.
+            return 200, {"error": "nodenames is empty; make sure the extender "
+                                  "config sets nodeCacheCapable=true"}, None
+        sch = self.registry.for_pod(pod)
+        if sch is None:
+            # Not a GPU pod: pass every node through unchanged.
+            return 200, {"nodenames": node_names, "failedNodes": {}}, None
+        metrics.TRACKER.saw_filter(obj.pod_uid(pod))
+        ok, failed = sch.assume(list(node_names), pod)
+        metrics.NODES_CACHED.set(len(sch.state.node_names()))
+        return 200, {"nodenames": ok, "failedNodes": failed}, None
+
+    def priorities(self, body: bytes):
+        args = _parse_json(body)
+        pod = args.get("pod")
+        if not pod:
+            raise _BadRequest("ExtenderArgs.pod missing")
+        node_names = args.get("nodenames") or []
+        sch = self.registry.for_pod(pod)
+        if sch is None:
+            result = [{"host": n, "score": 0} for n in node_names]
+            return 200, result, None
+        scores = sch.score(list(node_names), pod)
+        # Extender protocol: integer scores 0..10 before weighting.
+        result = [{"host": n, "score": int(round(s))}
+                  for n, s in zip(node_names, scores)]
+        return 200, result, None
+
+    def bind(self, body: bytes):
+        args = _parse_json(body)
+        name = args.get("podName", "")
+        ns = args.get("podNamespace", "default")
+        uid = args.get("podUID", "")
+        node = args.get("node", "")
+        if not name or not node:
+            raise _BadRequest("ExtenderBindingArgs requires podName and node")
+        sch = self.registry.default
+        try:
+            pod = sch.client.get_pod(ns, name)
+        except NotFoundError:
+            return 500, {"error": f"pod {ns}/{name} not found"}, None
+        # Double-get UID consistency check (reference GetPod, pod.go:110-131).
+        if uid and obj.pod_uid(pod) != uid:
+            return 500, {"error": f"pod {ns}/{name} UID changed (recreated?)"}, None
+        if obj.is_completed_pod(pod):
+            return 500, {"error": f"pod {ns}/{name} is already completed"}, None
+        pod_sch = self.registry.for_pod(pod) or sch
+        try:
+            pod_sch.bind(node, pod)
+        except BindError as exc:
+            return 500, {"error": str(exc)}, None
+        metrics.PODS_SCHEDULED.inc()
+        metrics.TRACKER.saw_bind(obj.pod_uid(pod))
+        return 200, {}, None
+
+    def status(self, body: bytes):
+        return 200, None, self.registry.status_json().encode()
+
+    def version(self, body: bytes):
+        return 200, {"version": __version__,
+                     "target": "MI355X (gfx950)",
+                     "policies": sorted({s.policy for s in
+                                         self.registry.schedulers.values()})}, None
+
+    def metrics(self, body: bytes):
+        return 200, None, metrics.render()
+
+    def healthz(self, body: bytes):
+        return 200, {"ok": True}, None
+
+    def debug_stacks(self, body: bytes):
+        frames = sys._current_frames()
+        out = []
+        for tid, frame in frames.items():
+            out.append(f"--- thread {tid} ---")
+            out.extend(line.rstrip() for line in traceback.format_stack(frame))
+        return 200, None, ("\n".join(out) + "\n").encode()
+
+
+class _BadRequest(Exception):
+    pass
+
+
+def _parse_json(body: bytes) -> Dict[str, Any]:
+    try:
+        out = json.loads(body or b"{}")
+    except json.JSONDecodeError as exc:
+        raise _BadRequest(f"malformed JSON: {exc}") from exc
+    if not isinstance(out, dict):
+        raise _BadRequest("JSON body must be an object")
+    return out
+
+
+def make_app(registry: SchedulerRegistry) -> ExtenderApp:
+    return ExtenderApp(registry)
