@@ -1,0 +1,165 @@
+"""MI355X device inventory discovery.
+
+The reference scheduler ecosystem delegates device discovery to a separate
+node agent (elastic-gpu-agent) built on NVML; this is the MI355X-native
+replacement. Discovery order:
+
+  1. the in-tree HIP probe (`_gpuprobe`) — authoritative when a GPU is
+     visible: exact HBM3E bytes, CU count, gfx arch, measured health;
+  2. the `amdsmi` python bindings (ROCm's supported NVML analogue);
+  3. `rocm-smi --showmeminfo vram --json` / `amd-smi` subprocess parsing;
+  4. torch.cuda device properties (PyTorch-ROCm);
+  5. none (CPU-only machine) -> empty inventory.
+
+Each card is reported as {"index", "name", "gcn_arch", "memory_bytes",
+"core": 100}; the scheduler's node annotation codec
+(k8s.objects.node_devices) consumes exactly this shape.
+"""
+from __future__ import annotations
+
+import json
+import logging
+import shutil
+import subprocess
+from typing import Any, Dict, List
+
+from elastic_gpu_scheduler_amd.utils import types as t
+
+log = logging.getLogger("egs.agent")
+
+
+def _via_gpuprobe() -> List[Dict[str, Any]]:
+    from elastic_gpu_scheduler_amd._native import gpuprobe_available, gpuprobe
+
+    if not gpuprobe_available():
+        return []
+    probe = gpuprobe()
+    cards = []
+    for info in probe.inventory():
+        cards.append({
+            "index": info["index"],
+            "name": info["name"],
+            "gcn_arch": info["gcn_arch"],
+            "memory_bytes": int(info["total_mem_bytes"]),
+            "compute_units": info["multi_processor_count"],
+            "core": t.GPU_CORE_EACH_CARD,
+            "source": "gpuprobe",
+        })
+    return cards
+
+
+def _via_amdsmi() -> List[Dict[str, Any]]:
+    try:
+        import amdsmi  # type: ignore
+    except ImportError:
+        return []
+    try:
+        amdsmi.amdsmi_init()
+        cards = []
+        for i, handle in enumerate(amdsmi.amdsmi_get_processor_handles()):
+            asic = amdsmi.amdsmi_get_gpu_asic_info(handle)
+            mem = amdsmi.amdsmi_get_gpu_memory_total(
+                handle, amdsmi.AmdSmiMemoryType.VRAM)
+            cards.append({
+                "index": i,
+                "name": asic.get("market_name", "AMD GPU"),
+                "gcn_arch": asic.get("target_graphics_version", ""),
+                "memory_bytes": int(mem),
+                "core": t.GPU_CORE_EACH_CARD,
+                "source": "amdsmi",
+            })
+        amdsmi.amdsmi_shut_down()
+        return cards
+    except Exception:
+        log.debug("amdsmi inventory failed", exc_info=True)
+        return []
+
+
+def parse_rocm_smi_vram(payload: str) -> List[Dict[str, Any]]:
+    """Parse `rocm-smi --showmeminfo vram --json` output into cards.
+
+    Shape: {"card0": {"VRAM Total Memory (B)": "309237645312", ...}, ...}
+    (exercised against canned output in tests — no GPU needed).
+    """
+    try:
+        data = json.loads(payload)
+    except json.JSONDecodeError:
+        return []
+    cards = []
+    for key in sorted(data.keys(), key=lambda k: (len(k), k)):
+        if not key.startswith("card"):
+            continue
+        entry = data[key]
+        total = None
+        for field in ("VRAM Total Memory (B)", "vram_total", "VRAM Total Memory"):
+            if field in entry:
+                try:
+                    total = int(str(entry[field]).strip())
+                except ValueError:
+                    total = None
+                break
+        if total is None:
+            continue
+        cards.append({
+            "index": int(key[len("card"):]),
+            "name": "AMD GPU",
+            "gcn_arch": "",
+            "memory_bytes": total,
+            "core": t.GPU_CORE_EACH_CARD,
+            "source": "rocm-smi",
+        })
+    return cards
+
+
+def _via_rocm_smi() -> List[Dict[str, Any]]:
+    exe = shutil.which("rocm-smi")
+    if not exe:
+        return []
+    try:
+        out = subprocess.run([exe, "--showmeminfo", "vram", "--json"],
+                             capture_output=True, text=True, timeout=30)
+        return parse_rocm_smi_vram(out.stdout)
+    except (subprocess.SubprocessError, OSError):
+        return []
+
+
+def _via_torch() -> List[Dict[str, Any]]:
+    try:
+        import torch
+    except ImportError:
+        return []
+    if not torch.cuda.is_available():
+        return []
+    cards = []
+    for i in range(torch.cuda.device_count()):
+        prop = torch.cuda.get_device_properties(i)
+        cards.append({
+            "index": i,
+            "name": prop.name,
+            "gcn_arch": getattr(prop, "gcnArchName", ""),
+            "memory_bytes": int(prop.total_memory),
+            "core": t.GPU_CORE_EACH_CARD,
+            "source": "torch",
+        })
+    return cards
+
+
+def discover(prefer: str = "auto") -> List[Dict[str, Any]]:
+    """Discover the node's GPU cards. `prefer` forces one source in tests."""
+    sources = {
+        "gpuprobe": _via_gpuprobe,
+        "amdsmi": _via_amdsmi,
+        "rocm-smi": _via_rocm_smi,
+        "torch": _via_torch,
+    }
+    if prefer != "auto":
+        return sources[prefer]()
+    for name in ("gpuprobe", "amdsmi", "rocm-smi", "torch"):
+        try:
+            cards = sources[name]()
+        except Exception:
+            log.debug("inventory source %s failed", name, exc_info=True)
+            cards = []
+        if cards:
+            return cards
+    return []

@@ -1,0 +1,104 @@
+"""xGMI topology discovery for the node agent.
+
+Produces the hop matrix the scheduler's Topology consumes:
+  0 = same card, 1 = direct xGMI link, 2+ = routed (multi-hop / PCIe / CPU).
+
+Sources, in order:
+  1. the HIP probe's peer-access + performance-rank matrix (authoritative on
+     a live box);
+  2. `rocm-smi --showtopohops` parsing (works without HIP context);
+  3. default: fully connected single hive — correct for a standard
+     8x MI355X OAM board, where every card pair shares one of the 7
+     point-to-point xGMI links.
+"""
+from __future__ import annotations
+
+import logging
+import re
+import shutil
+import subprocess
+from typing import List
+
+log = logging.getLogger("egs.agent")
+
+
+def default_hive(n: int) -> List[List[int]]:
+    return [[0 if i == j else 1 for j in range(n)] for i in range(n)]
+
+
+def _via_gpuprobe() -> List[List[int]]:
+    from elastic_gpu_scheduler_amd._native import gpuprobe_available, gpuprobe
+
+    if not gpuprobe_available():
+        return []
+    probe = gpuprobe()
+    if probe.device_count() == 0:
+        return []
+    return [list(row) for row in probe.xgmi_hop_matrix()]
+
+
+def parse_showtopohops(text: str) -> List[List[int]]:
+    """Parse `rocm-smi --showtopohops` tabular output.
+
+    Expected shape (one header row + one row per GPU):
+             GPU0  GPU1  GPU2
+      GPU0   0     1     1
+      GPU1   1     0     1
+      GPU2   1     1     0
+    Exercised against canned output in tests.
+    """
+    rows = {}
+    header_cols: List[int] = []
+    for line in text.splitlines():
+        line = line.strip()
+        m = re.match(r"^GPU(\d+)\s+(.*)$", line)
+        if not m:
+            if re.match(r"^(GPU\d+\s*)+$", line):
+                header_cols = [int(x) for x in re.findall(r"GPU(\d+)", line)]
+            continue
+        gpu = int(m.group(1))
+        vals = []
+        for tok in m.group(2).split():
+            try:
+                vals.append(int(float(tok)))
+            except ValueError:
+                vals.append(-1)
+        rows[gpu] = vals
+    if not rows:
+        return []
+    n = max(rows.keys()) + 1
+    matrix = [[0] * n for _ in range(n)]
+    for i in range(n):
+        vals = rows.get(i, [])
+        for col, v in enumerate(vals[:n]):
+            j = header_cols[col] if col < len(header_cols) else col
+            matrix[i][j] = max(v, 0)
+    return matrix
+
+
+def _via_rocm_smi() -> List[List[int]]:
+    exe = shutil.which("rocm-smi")
+    if not exe:
+        return []
+    try:
+        out = subprocess.run([exe, "--showtopohops"], capture_output=True,
+                             text=True, timeout=30)
+        return parse_showtopohops(out.stdout)
+    except (subprocess.SubprocessError, OSError):
+        return []
+
+
+def discover(n_cards: int, prefer: str = "auto") -> List[List[int]]:
+    sources = {"gpuprobe": _via_gpuprobe, "rocm-smi": _via_rocm_smi}
+    if prefer in sources:
+        m = sources[prefer]()
+        return m if m else default_hive(n_cards)
+    for name in ("gpuprobe", "rocm-smi"):
+        try:
+            m = sources[name]()
+        except Exception:
+            log.debug("topology source %s failed", name, exc_info=True)
+            m = []
+        if m:
+            return m
+    return default_hive(n_cards)

@@ -125,6 +125,8 @@ PYBIND11_MODULE(_core, m) {
       .def("add_pod", &ClusterState::add_pod, py::arg("node"), py::arg("uid"),
            py::arg("request"), py::arg("option"),
            py::call_guard<py::gil_scoped_release>())
+      .def("note_pod_node", &ClusterState::note_pod_node, py::arg("uid"),
+           py::arg("node"), py::call_guard<py::gil_scoped_release>())
       .def("forget_pod", &ClusterState::forget_pod,
            py::call_guard<py::gil_scoped_release>())
       .def("known_pod", &ClusterState::known_pod,

@@ -1,0 +1,138 @@
+"""Concurrency: the properties that justify per-node locking — no
+double-booking under concurrent binds, parallel filter fan-out, and
+linearizable assume/bind per node (SURVEY.md hard part #2)."""
+from __future__ import annotations
+
+import threading
+
+from elastic_gpu_scheduler_amd._native import core
+from elastic_gpu_scheduler_amd.k8s.client import FakeKubeClient
+from elastic_gpu_scheduler_amd.scheduler.service import BindError, GPUUnitScheduler
+from tests.conftest import make_node, make_pod
+
+GiB = 1024**3
+
+
+def test_no_double_booking_under_concurrent_allocate():
+    """64 threads race to allocate 60% of a 2-card node; capacity admits
+    exactly 2 (+1 on the second card? no: 60% twice doesn't fit one card).
+    Total admitted must never exceed what fits."""
+    c = core.ClusterState("binpack", 0, 8)
+    c.add_node("n", [core.Device(100, 100, 288 * GiB, 288 * GiB)
+                     for _ in range(2)], [])
+    admitted = []
+    lock = threading.Lock()
+
+    def worker(i):
+        try:
+            c.allocate("n", f"p{i}", [core.GPUUnit(0, 60, GiB)])
+            with lock:
+                admitted.append(i)
+        except RuntimeError:
+            pass
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(64)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert len(admitted) == 2  # one 60% per card, no more
+    devs = c.node_devices("n")
+    assert all(d.core_avail >= 0 and d.mem_avail >= 0 for d in devs)
+
+
+def test_concurrent_whole_card_allocations_never_overlap():
+    c = core.ClusterState("binpack", 0, 8)
+    c.add_node("n", [core.Device(100, 100, 288 * GiB, 288 * GiB)
+                     for _ in range(8)], [])
+    placements = {}
+    lock = threading.Lock()
+
+    def worker(i):
+        try:
+            opt = c.allocate("n", f"p{i}", [core.GPUUnit(2, 0, 0)])
+            with lock:
+                placements[i] = opt.allocated[0]
+        except RuntimeError:
+            pass
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert len(placements) == 4  # 8 cards / 2 per pod
+    used = [idx for cards in placements.values() for idx in cards]
+    assert len(used) == len(set(used)) == 8
+
+
+def test_parallel_assume_across_many_nodes():
+    c = core.ClusterState("binpack", 0, 0)
+    names = [f"n{i}" for i in range(64)]
+    for n in names:
+        c.add_node(n, [core.Device(100, 100, 288 * GiB, 288 * GiB)
+                       for _ in range(8)], [])
+    verdicts = c.assume(names, "p", [core.GPUUnit(0, 50, 64 * GiB)])
+    assert verdicts == [0] * 64
+    scores = c.score(names, "p", [core.GPUUnit(0, 50, 64 * GiB)])
+    assert len(scores) == 64
+
+
+def test_concurrent_bind_through_service_no_overcommit():
+    client = FakeKubeClient()
+    client.add_node(make_node("n1", cards=1))
+    sch = GPUUnitScheduler(client, threads=8)
+    pods = [client.create_pod(make_pod(f"p{i}", core=30, memory=10 * GiB))
+            for i in range(8)]
+    for p in pods:
+        sch.assume(["n1"], p)
+    results = {}
+    lock = threading.Lock()
+
+    def worker(p):
+        try:
+            sch.bind("n1", p)
+            with lock:
+                results[p["metadata"]["name"]] = "ok"
+        except (BindError, RuntimeError) as exc:
+            with lock:
+                results[p["metadata"]["name"]] = "rejected"
+
+    threads = [threading.Thread(target=worker, args=(p,)) for p in pods]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    ok = [k for k, v in results.items() if v == "ok"]
+    assert len(ok) == 3  # 3 x 30% fits one card, 4th does not
+    d = sch.state.node_devices("n1")[0]
+    assert d.core_avail == 10
+    assert d.mem_avail == (288 - 30) * GiB
+
+
+def test_forget_while_assume_storm():
+    """Interleaved assume/allocate/forget across threads keeps accounting
+    consistent (ends balanced at zero usage)."""
+    c = core.ClusterState("binpack", 0, 4)
+    c.add_node("n", [core.Device(100, 100, 288 * GiB, 288 * GiB)
+                     for _ in range(8)], [])
+
+    def cycle(i):
+        uid = f"p{i}"
+        req = [core.GPUUnit(0, 10 + (i % 5) * 10, GiB)]
+        for _ in range(20):
+            c.assume(["n"], uid, req)
+            try:
+                c.allocate("n", uid, req)
+            except RuntimeError:
+                continue
+            c.forget_pod(uid)
+
+    threads = [threading.Thread(target=cycle, args=(i,)) for i in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    # nothing left allocated
+    devs = c.node_devices("n")
+    assert all(d.core_avail == 100 and d.mem_avail == 288 * GiB for d in devs)

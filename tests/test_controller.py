@@ -1,0 +1,126 @@
+"""Reconcile controller: watch events keep accounting honest (reference
+pkg/controller/controller.go syncPod semantics)."""
+from __future__ import annotations
+
+import time
+
+from elastic_gpu_scheduler_amd.controller.controller import Controller
+from elastic_gpu_scheduler_amd.k8s import objects as obj
+from elastic_gpu_scheduler_amd.k8s.client import FakeKubeClient
+from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+from tests.conftest import make_node, make_pod
+
+GiB = 1024**3
+
+
+def wait_until(pred, timeout=5.0):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        try:
+            if pred():
+                return True
+        except RuntimeError:
+            pass  # e.g. node cache not yet filled by the controller
+        time.sleep(0.01)
+    return False
+
+
+def make_stack():
+    client = FakeKubeClient()
+    client.add_node(make_node("n1"))
+    registry = SchedulerRegistry(client)
+    ctrl = Controller(client, registry, workers=2, resync_seconds=3600)
+    ctrl.start()
+    return client, registry, ctrl
+
+
+def test_completed_pod_is_released():
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=40, memory=GiB))
+        sch.assume(["n1"], pod)
+        sch.bind("n1", pod)
+        assert any(d.core_avail == 60 for d in sch.state.node_devices("n1"))
+        client.set_pod_phase("default", "p", "Succeeded")
+        assert wait_until(lambda: all(
+            d.core_avail == 100 for d in sch.state.node_devices("n1")))
+    finally:
+        ctrl.stop()
+
+
+def test_deleted_pod_is_released():
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=40, memory=GiB))
+        sch.assume(["n1"], pod)
+        sch.bind("n1", pod)
+        client.delete_pod("default", "p")
+        assert wait_until(lambda: all(
+            d.core_avail == 100 for d in sch.state.node_devices("n1")))
+        # late MODIFIED must not resurrect (tombstone)
+        sch.add_pod(client_pod_with_node(pod))
+        assert all(d.core_avail == 100 for d in sch.state.node_devices("n1"))
+    finally:
+        ctrl.stop()
+
+
+def client_pod_with_node(pod):
+    p = dict(pod)
+    p["spec"] = dict(pod["spec"], nodeName="n1")
+    return p
+
+
+def test_externally_assigned_pod_is_accounted():
+    """A pod bound by another scheduler replica / before our startup shows
+    up via the watch and must be charged (reference assignPod path,
+    controller.go:174-180)."""
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = make_pod("ext", core=30, memory=GiB)
+        pod["metadata"]["labels"] = {"elasticgpu.io/assumed": "true"}
+        pod["metadata"]["annotations"] = {
+            "elasticgpu.io/assumed": "true",
+            "elasticgpu.io/container-c0": "0",
+        }
+        pod["spec"]["nodeName"] = "n1"
+        client.create_pod(pod)
+        assert wait_until(lambda: any(
+            d.core_avail == 70 for d in sch.state.node_devices("n1")))
+    finally:
+        ctrl.stop()
+
+
+def test_resync_evicts_vanished_pods():
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=40, memory=GiB))
+        sch.assume(["n1"], pod)
+        sch.bind("n1", pod)
+        # vanish without a DELETE event: remove directly from the store
+        with client._mu:
+            client._pods.clear()
+        ctrl.resync_once()
+        assert all(d.core_avail == 100 for d in sch.state.node_devices("n1"))
+    finally:
+        ctrl.stop()
+
+
+def test_non_gpu_pods_ignored():
+    client, registry, ctrl = make_stack()
+    try:
+        client.create_pod({
+            "metadata": {"name": "cpu", "namespace": "default"},
+            "spec": {"containers": [{"name": "c",
+                                     "resources": {"requests": {"cpu": "1"}}}],
+                     "nodeName": "n1"},
+            "status": {"phase": "Running"},
+        })
+        time.sleep(0.1)
+        # never even cached the node: non-GPU pods are filtered at the watch
+        assert not registry.default.state.has_node("n1")
+    finally:
+        ctrl.stop()

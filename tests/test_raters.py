@@ -1,0 +1,96 @@
+"""Scoring policies: binpack packs, spread spreads, random is calibrated and
+deterministic, topology steers multi-card placements onto linked hives."""
+from __future__ import annotations
+
+from elastic_gpu_scheduler_amd._native import core
+
+GiB = 1024**3
+
+
+def devices(n=8):
+    return [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(n)]
+
+
+def frac(core_pct, mem=0):
+    return [core.GPUUnit(0, core_pct, mem)]
+
+
+def test_binpack_prefers_partially_used_card():
+    c = core.ClusterState("binpack", 0, 0)
+    c.add_node("n", devices(4), [])
+    c.allocate("n", "warm", frac(50, 50 * GiB))
+    warm_card = c.node_devices("n")
+    used = [i for i, d in enumerate(warm_card) if d.core_avail < 100][0]
+    opt = c.allocate("n", "p", frac(30, 10 * GiB))
+    assert opt.allocated == [[used]]
+
+
+def test_spread_prefers_empty_card():
+    c = core.ClusterState("spread", 0, 0)
+    c.add_node("n", devices(4), [])
+    c.allocate("n", "warm", frac(50, 50 * GiB))
+    used = [i for i, d in enumerate(c.node_devices("n")) if d.core_avail < 100][0]
+    opt = c.allocate("n", "p", frac(30, 10 * GiB))
+    assert opt.allocated[0][0] != used
+
+
+def test_spread_is_not_a_stub():
+    """Reference Spread always returns 0 (rater.go:56-59); ours must produce
+    differentiated node scores."""
+    c = core.ClusterState("spread", 0, 0)
+    c.add_node("empty", devices(2), [])
+    c.add_node("busy", devices(2), [])
+    c.allocate("busy", "w1", frac(90, 200 * GiB))
+    c.allocate("busy", "w2", frac(90, 200 * GiB))
+    scores = c.score(["empty", "busy"], "p", frac(10, GiB))
+    assert scores[0] > scores[1]
+
+
+def test_scores_calibrated_zero_to_ten():
+    for policy in ("binpack", "spread", "random"):
+        c = core.ClusterState(policy, 3, 0)
+        c.add_node("n", devices(4), [])
+        for i in range(6):
+            s = c.score(["n"], f"p{i}", frac(10 + i * 5, i * GiB))[0]
+            assert 0.0 <= s <= 10.0, (policy, s)
+
+
+def test_random_varies_across_pods_but_stable_per_pod():
+    c = core.ClusterState("random", 42, 0)
+    for name in ("a", "b", "c", "d"):
+        c.add_node(name, devices(2), [])
+    nodes = ["a", "b", "c", "d"]
+    s1 = c.score(nodes, "p1", frac(10, GiB))
+    s1_again = c.score(nodes, "p1", frac(10, GiB))
+    assert s1 == s1_again
+    # different nodes should not all score identically (salted per node)
+    assert len({round(x, 6) for x in s1}) > 1
+
+
+def test_topology_prefers_linked_hive():
+    """gpu-core=400 must land on 4 xGMI-adjacent cards (BASELINE config #5):
+    cards 0-3 and 4-7 are separate hives (1 hop inside, 3 across)."""
+    hops = [[0 if i == j else (1 if (i < 4) == (j < 4) else 3)
+             for j in range(8)] for i in range(8)]
+    for policy in ("binpack", "spread", "random"):
+        c = core.ClusterState(policy, 0, 0)
+        c.add_node("n", devices(8), hops)
+        opt = c.allocate("n", "p", [core.GPUUnit(4, 0, 0)])
+        cards = opt.allocated[0]
+        assert len(cards) == 4
+        sides = {i < 4 for i in cards}
+        assert len(sides) == 1, f"{policy}: crossed hives: {cards}"
+
+
+def test_topology_locality_values():
+    hops = [[0 if i == j else (1 if (i < 2) == (j < 2) else 3)
+             for j in range(4)] for i in range(4)]
+    topo = core.Topology(hops)
+    assert topo.locality([0, 1]) == 1.0
+    assert topo.locality([0, 2]) < topo.locality([0, 1])
+    assert topo.set_cost([0, 1, 2, 3]) == 2 * 1 + 4 * 3
+
+
+def test_uniform_topology_locality_is_one():
+    topo = core.Topology([])
+    assert topo.locality([0, 5]) == 1.0

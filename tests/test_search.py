@@ -1,0 +1,131 @@
+"""Placement search: correctness vs a Python brute-force oracle on small
+configurations, plus budget/pruning behavior."""
+from __future__ import annotations
+
+import itertools
+
+from elastic_gpu_scheduler_amd._native import core
+
+GiB = 1024**3
+
+
+def brute_force_feasible(avail, req):
+    """Oracle: does ANY assignment exist? avail = [(core, mem)], req =
+    [(kind, a, b)] with kind 'whole' (a=count) or 'frac' (a=core, b=mem)."""
+    n = len(avail)
+
+    def rec(state, i):
+        if i == len(req):
+            return True
+        kind, a, b = req[i]
+        if kind == "none":
+            return rec(state, i + 1)
+        if kind == "whole":
+            free = [j for j in range(n)
+                    if state[j] == avail[j] and state[j][0] == 100]
+            for combo in itertools.combinations(free, a):
+                s2 = list(state)
+                for j in combo:
+                    s2[j] = (0, 0)
+                if rec(s2, i + 1):
+                    return True
+            return False
+        for j in range(n):
+            cj, mj = state[j]
+            if cj >= a and mj >= b:
+                s2 = list(state)
+                s2[j] = (cj - a, mj - b)
+                if rec(s2, i + 1):
+                    return True
+        return False
+
+    return rec(list(avail), 0)
+
+
+def to_native(avail, req):
+    devs = []
+    for c, m in avail:
+        devs.append(core.Device(100, c, 288 * GiB, m))
+    units = []
+    for kind, a, b in req:
+        if kind == "whole":
+            units.append(core.GPUUnit(a, 0, 0))
+        elif kind == "frac":
+            units.append(core.GPUUnit(0, a, b))
+        else:
+            units.append(core.GPUUnit(0, 0, 0))
+    return devs, units
+
+
+CASES = [
+    # (avail per card, request per container)
+    ([(100, 288 * GiB)] * 4, [("frac", 30, 10 * GiB)]),
+    ([(100, 288 * GiB)] * 4, [("whole", 2, 0), ("frac", 50, GiB)]),
+    ([(50, 100 * GiB), (100, 288 * GiB)], [("whole", 1, 0)]),
+    ([(50, 100 * GiB), (40, 20 * GiB)], [("frac", 45, 50 * GiB)]),
+    ([(10, GiB)] * 3, [("frac", 20, 0)]),                      # infeasible
+    ([(100, 288 * GiB)] * 2, [("whole", 3, 0)]),               # infeasible
+    ([(100, 288 * GiB)] * 3,
+     [("frac", 60, 0), ("frac", 60, 0), ("frac", 60, 0), ("frac", 60, 0)]),
+    ([(100, 288 * GiB)] * 2, [("none", 0, 0), ("frac", 10, GiB)]),
+    ([(30, 10 * GiB), (30, 10 * GiB), (100, 288 * GiB)],
+     [("frac", 25, 5 * GiB), ("whole", 1, 0)]),
+    ([(100, 64 * GiB)] * 4, [("frac", 0, 65 * GiB)]),          # mem infeasible
+]
+
+
+def test_search_matches_oracle_feasibility():
+    for avail, req in CASES:
+        devs, units = to_native(avail, req)
+        for policy in ("binpack", "spread"):
+            feasible, opt, _ = core.search_placement(devs, units, policy, 0, [])
+            assert feasible == brute_force_feasible(avail, req), (avail, req, policy)
+
+
+def test_search_respects_capacity():
+    """Applying the returned option must never overdraw any card."""
+    for avail, req in CASES:
+        devs, units = to_native(avail, req)
+        feasible, opt, _ = core.search_placement(devs, units, "binpack", 0, [])
+        if not feasible:
+            continue
+        use = [[0, 0] for _ in avail]
+        for c, cards in enumerate(opt.allocated):
+            kind, a, b = req[c]
+            for j in cards:
+                if kind == "whole":
+                    use[j][0] += 100
+                    use[j][1] += avail[j][1]
+                else:
+                    use[j][0] += a
+                    use[j][1] += b
+        for j, (c_used, m_used) in enumerate(use):
+            assert c_used <= avail[j][0]
+            assert m_used <= avail[j][1]
+
+
+def test_leaf_budget_bounds_work():
+    """A worst-case shape (many containers x many cards) stays within the
+    deterministic leaf budget instead of exploding combinatorially."""
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(16)]
+    units = [core.GPUUnit(0, 5, GiB) for _ in range(10)]
+    feasible, opt, leaves = core.search_placement(devs, units, "binpack", 0, [])
+    assert feasible
+    assert leaves <= 4096
+
+
+def test_whole_card_picks_topology_best_not_first_free():
+    """With card 0 free but isolated, and cards 2,3 linked, a 2-card pod must
+    take the linked pair (the reference would take the first 2 free:
+    gpu.go:96-108)."""
+    hops = [
+        [0, 3, 3, 3],
+        [3, 0, 3, 3],
+        [3, 3, 0, 1],
+        [3, 3, 1, 0],
+    ]
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(4)]
+    units = [core.GPUUnit(2, 0, 0)]
+    feasible, opt, _ = core.search_placement(devs, units, "binpack", 0, hops)
+    assert feasible
+    assert sorted(opt.allocated[0]) == [2, 3]
