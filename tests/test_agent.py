@@ -1,0 +1,114 @@
+"""Node agent: inventory/topology parsing from canned tool output (CPU),
+publish flow against the fake apiserver."""
+from __future__ import annotations
+
+import json
+
+from elastic_gpu_scheduler_amd.agent import inventory as inv
+from elastic_gpu_scheduler_amd.agent import topology as topo
+from elastic_gpu_scheduler_amd.agent.agent import NodeAgent
+from elastic_gpu_scheduler_amd.k8s import objects as obj
+from elastic_gpu_scheduler_amd.k8s.client import FakeKubeClient
+
+GiB = 1024**3
+
+ROCM_SMI_VRAM = json.dumps({
+    "card0": {"VRAM Total Memory (B)": "309237645312",
+              "VRAM Total Used Memory (B)": "12345"},
+    "card1": {"VRAM Total Memory (B)": "309237645312",
+              "VRAM Total Used Memory (B)": "0"},
+})
+
+SHOWTOPOHOPS = """
+========================= ROCm System Management Interface =========================
+=========================== Hops between two GPUs ==================================
+       GPU0   GPU1   GPU2   GPU3
+GPU0   0      1      1      2
+GPU1   1      0      1      2
+GPU2   1      1      0      1
+GPU3   2      2      1      0
+====================================================================================
+"""
+
+
+def test_parse_rocm_smi_vram():
+    cards = inv.parse_rocm_smi_vram(ROCM_SMI_VRAM)
+    assert len(cards) == 2
+    assert cards[0]["index"] == 0
+    assert cards[0]["memory_bytes"] == 309237645312
+    assert cards[0]["core"] == 100
+
+
+def test_parse_rocm_smi_vram_garbage():
+    assert inv.parse_rocm_smi_vram("not json") == []
+    assert inv.parse_rocm_smi_vram("{}") == []
+
+
+def test_parse_showtopohops():
+    m = topo.parse_showtopohops(SHOWTOPOHOPS)
+    assert len(m) == 4
+    assert m[0][0] == 0
+    assert m[0][1] == 1
+    assert m[0][3] == 2
+    assert m[3][2] == 1
+
+
+def test_parse_showtopohops_garbage():
+    assert topo.parse_showtopohops("") == []
+    assert topo.parse_showtopohops("random\ntext") == []
+
+
+def test_default_hive():
+    m = topo.default_hive(8)
+    assert m[0][0] == 0 and m[0][7] == 1 and len(m) == 8
+
+
+def test_agent_publish_roundtrips_into_scheduler_inventory(monkeypatch):
+    """agent.publish() -> node annotations -> scheduler node_devices must
+    agree with what the agent discovered."""
+    cards = [{"index": i, "name": "MI355X", "gcn_arch": "gfx950",
+              "memory_bytes": 288 * GiB, "core": 100, "source": "test"}
+             for i in range(8)]
+    monkeypatch.setattr(inv, "discover", lambda prefer="auto": cards)
+    monkeypatch.setattr(topo, "discover",
+                        lambda n, prefer="auto": topo.default_hive(n))
+
+    client = FakeKubeClient()
+    client.add_node({"metadata": {"name": "gpu-node"}, "status": {}})
+    agent = NodeAgent("gpu-node", client)
+    ann = agent.publish()
+    node = client.get_node("gpu-node")
+    assert node["metadata"]["annotations"] == ann
+
+    devs = obj.node_devices(node)
+    assert len(devs) == 8
+    assert devs[0].mem_total == 288 * GiB
+    hops = obj.node_topology(node)
+    assert len(hops) == 8 and hops[0][1] == 1
+
+
+def test_agent_allocatable(monkeypatch):
+    cards = [{"index": i, "memory_bytes": 288 * GiB, "core": 100}
+             for i in range(8)]
+    monkeypatch.setattr(inv, "discover", lambda prefer="auto": cards)
+    monkeypatch.setattr(topo, "discover",
+                        lambda n, prefer="auto": topo.default_hive(n))
+    agent = NodeAgent("n")
+    alloc = agent.allocatable()
+    assert alloc["elasticgpu.io/gpu-core"] == "800"
+    assert alloc["elasticgpu.io/gpu-memory"] == str(8 * 288 * GiB)
+    assert alloc["amd.com/gpu"] == "8"
+
+    node = agent.node_object()
+    devs = obj.node_devices(node)
+    assert len(devs) == 8
+
+
+def test_discover_empty_on_cpu_only_box():
+    # On this CPU-only container every real source returns [] -> empty.
+    import torch
+
+    if torch.cuda.is_available():
+        return  # running on a GPU box: covered by gpu-marked tests
+    assert inv.discover() == []
+    assert topo.discover(8) == topo.default_hive(8)
