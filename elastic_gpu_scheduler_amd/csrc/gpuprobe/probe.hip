@@ -24,6 +24,7 @@
 #include <cstdint>
 #include <stdexcept>
 #include <string>
+#include <utility>
 #include <vector>
 
 namespace py = pybind11;
@@ -71,6 +72,13 @@ int checked_device_count() {
   hipError_t e = hipGetDeviceCount(&n);
   if (e != hipSuccess) return 0;
   return n;
+}
+
+// Diagnostic: raw device-count call result (count, error string).
+std::pair<int, std::string> hip_status() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  return {e == hipSuccess ? n : -1, hipGetErrorString(e)};
 }
 
 void require_device(int device) {
@@ -252,6 +260,10 @@ static bool stamp(int device, uint64_t tag, int mib) {
 PYBIND11_MODULE(_gpuprobe, m) {
   m.doc() = "MI355X (gfx950) device probe: inventory, HBM/xGMI bandwidth, placement stamp";
   m.def("device_count", &checked_device_count);
+  m.def("hip_status", [] {
+    auto s = hip_status();
+    return py::make_tuple(s.first, s.second);
+  });
   m.def("device_info", &device_info, py::arg("device"));
   m.def("inventory", &inventory);
   m.def("hbm_bandwidth", &hbm_bandwidth, py::arg("device") = 0, py::arg("mib") = 256,

@@ -136,9 +136,9 @@ def allocation_annotations(pod: Pod, allocated: List[List[int]],
 def apply_allocation(pod: Pod, allocated: List[List[int]], node: str = "",
                      score: float = 0.0) -> Pod:
     """Return a copy of the pod with placement annotations + assumed label."""
-    import copy
+    from elastic_gpu_scheduler_amd.k8s.client import _jcopy
 
-    p = copy.deepcopy(pod)
+    p = _jcopy(pod)
     meta = p.setdefault("metadata", {})
     ann = meta.setdefault("annotations", {})
     ann.update(allocation_annotations(pod, allocated, node, score))

@@ -53,6 +53,9 @@ class GPUUnitScheduler:
         # a DELETE seen before the final MODIFIED must not re-add the pod.
         self._released: Dict[str, float] = {}
         self._released_mu = threading.Lock()
+        # fast-path membership cache over state's node map (adds only; a
+        # removed node falls back to the authoritative native check)
+        self._known_nodes: set = set()
         self.warm_start()
 
     # ---- cache management ------------------------------------------------
@@ -80,7 +83,10 @@ class GPUUnitScheduler:
     def _ensure_node(self, node_name: str, replay: bool = True) -> bool:
         """Lazily fill the node cache from a live Get + assumed-pod replay
         (reference getNodeInfo, scheduler.go:62-84)."""
+        if node_name in self._known_nodes:
+            return True
         if self.state.has_node(node_name):
+            self._known_nodes.add(node_name)
             return True
         try:
             node = self.client.get_node(node_name)
@@ -94,6 +100,7 @@ class GPUUnitScheduler:
             return False
         topo = obj.node_topology(node)
         self.state.add_node(node_name, devices, topo)
+        self._known_nodes.add(node_name)
         if replay:
             try:
                 pods = self.client.list_pods(
@@ -105,6 +112,12 @@ class GPUUnitScheduler:
                 if not obj.is_completed_pod(pod):
                     self._replay_pod(node_name, pod)
         return True
+
+    def invalidate_node(self, node_name: str) -> None:
+        """Evict a node from the cache so fresh inventory/topology (e.g. the
+        agent republishing annotations) is re-read on next use."""
+        self._known_nodes.discard(node_name)
+        self.state.remove_node(node_name)
 
     def _replay_pod(self, node_name: str, pod: Dict[str, Any]) -> None:
         allocated = obj.parse_allocation(pod)

@@ -68,7 +68,7 @@ def test_lazy_node_fill_replays_existing_pods():
     sch1.bind("n1", client.get_pod("default", "p"))
 
     sch2 = GPUUnitScheduler(client)
-    sch2.state.remove_node("n1")  # force a cold cache for the lazy-fill path
+    sch2.invalidate_node("n1")  # force a cold cache for the lazy-fill path
     ok, failed = sch2.assume(["n1"], client.create_pod(make_pod("q", core=80)))
     assert ok == ["n1"]
     # _ensure_node replayed the assumed pod: 25 core already used on one card
