@@ -28,7 +28,6 @@ For N>1 the driver launches this under torch.distributed.run.
 from __future__ import annotations
 
 import argparse
-import asyncio
 import json
 import os
 import socket
@@ -108,6 +107,51 @@ def build_cluster(n_nodes: int, cards: int, use_gpu_inventory: bool):
 
 # ---------------------------------------------------------------- pipeline
 
+class MiniHttpClient:
+    """Minimal blocking HTTP/1.1 keep-alive client over one socket."""
+
+    def __init__(self, host: str, port: int):
+        self.sock = socket.create_connection((host, port))
+        self.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        self.buf = b""
+
+    def close(self):
+        try:
+            self.sock.close()
+        except OSError:
+            pass
+
+    def post_json(self, path: str, payload) -> "tuple[int, object]":
+        body = json.dumps(payload).encode()
+        req = (f"POST {path} HTTP/1.1\r\nhost: bench\r\n"
+               f"content-type: application/json\r\n"
+               f"content-length: {len(body)}\r\n\r\n").encode() + body
+        self.sock.sendall(req)
+        # read headers
+        while b"\r\n\r\n" not in self.buf:
+            chunk = self.sock.recv(65536)
+            if not chunk:
+                raise ConnectionError("server closed connection")
+            self.buf += chunk
+        head, _, rest = self.buf.partition(b"\r\n\r\n")
+        lines = head.split(b"\r\n")
+        status = int(lines[0].split()[1])
+        clen = 0
+        for line in lines[1:]:
+            k, _, v = line.partition(b":")
+            if k.strip().lower() == b"content-length":
+                clen = int(v.strip())
+        while len(rest) < clen:
+            chunk = self.sock.recv(65536)
+            if not chunk:
+                raise ConnectionError("server closed connection")
+            rest += chunk
+        self.buf = rest[clen:]
+        return status, json.loads(rest[:clen] or b"{}")
+
+
+
+
 class BenchPipeline:
     def __init__(self, rank: int, args, device_index: int, use_gpu: bool):
         from elastic_gpu_scheduler_amd.controller.controller import Controller
@@ -127,9 +171,12 @@ class BenchPipeline:
         self.app = make_app(self.registry)
         self.node_names = [f"node-{i}" for i in range(args.nodes)]
         self.latencies: list[float] = []
+        self._lat_mu = threading.Lock()
         self.bind_retries = 0
+        self.port = None
         self._server = None
         self._server_thread = None
+        self._native = None
         self.base_url = None
         if not args.no_http:
             self._start_server()
@@ -141,6 +188,14 @@ class BenchPipeline:
             self.probe = None
 
     def _start_server(self):
+        if self.args.server == "native":
+            from elastic_gpu_scheduler_amd.server.native import NativeFrontend
+
+            self._native = NativeFrontend(self.app, host="127.0.0.1", port=0)
+            self._native.start()
+            self.port = self._native.port
+            self.base_url = f"http://127.0.0.1:{self.port}"
+            return
         import uvicorn
 
         port = self._free_port()
@@ -150,6 +205,7 @@ class BenchPipeline:
         self._server_thread = threading.Thread(target=self._server.run,
                                                daemon=True)
         self._server_thread.start()
+        self.port = port
         self.base_url = f"http://127.0.0.1:{port}"
         deadline = time.time() + 30
         import httpx
@@ -179,58 +235,74 @@ class BenchPipeline:
         if self.args.no_http:
             self._schedule_direct(pods, record_latency)
         else:
-            asyncio.run(self._schedule_http(pods, record_latency))
+            self._schedule_http(pods, record_latency)
         self._verify_sample(pods)
         self._release(pods)
 
-    async def _schedule_http(self, pods, record_latency: bool):
-        import aiohttp
+    def _schedule_http(self, pods, record_latency: bool):
+        """Threaded load generator: each worker holds one keep-alive HTTP
+        connection (kube-scheduler style) and drives pods through
+        filter -> priorities -> bind, retrying a failed bind like the real
+        scheduler requeues a pod. Raw sockets keep client overhead ~tens of
+        us so the SERVER is what gets measured."""
+        import queue as _queue
 
-        conn = aiohttp.TCPConnector(limit=self.args.concurrency)
-        async with aiohttp.ClientSession(
-                base_url=self.base_url, connector=conn,
-                timeout=aiohttp.ClientTimeout(total=60)) as c:
-            sem = asyncio.Semaphore(self.args.concurrency)
+        q: "_queue.Queue" = _queue.Queue()
+        for pod in pods:
+            q.put(pod)
+        errors = []
+        n_workers = min(self.args.concurrency, len(pods))
 
-            async def one(pod):
-                # Mimics kube-scheduler: ties between top-scored nodes break
-                # pseudo-randomly, and a failed bind re-enters the queue
-                # (filter -> priorities -> bind again).
-                async with sem:
-                    t0 = time.perf_counter()
-                    for attempt in range(8):
-                        async with c.post(
-                                "/scheduler/filter",
-                                json={"pod": pod,
-                                      "nodenames": self.node_names}) as r:
-                            body = await r.json()
-                        ok = body.get("nodenames") or []
-                        if not ok:
-                            raise RuntimeError(f"no feasible node: {body}")
-                        async with c.post(
-                                "/scheduler/priorities",
-                                json={"pod": pod, "nodenames": ok}) as r:
-                            prio = await r.json()
-                        top = max(e["score"] for e in prio)
-                        tied = [e["host"] for e in prio if e["score"] == top]
-                        best = tied[(hash(pod["metadata"]["uid"]) + attempt)
-                                    % len(tied)]
-                        async with c.post("/scheduler/bind", json={
-                                "podName": pod["metadata"]["name"],
-                                "podNamespace": "default",
-                                "podUID": pod["metadata"]["uid"],
-                                "node": best}) as r:
-                            status = r.status
-                            err = await r.text() if status != 200 else ""
-                        if status == 200:
-                            break
-                        self.bind_retries += 1
-                    else:
-                        raise RuntimeError(f"bind kept failing: {err}")
-                    if record_latency:
-                        self.latencies.append(time.perf_counter() - t0)
+        def worker():
+            conn = MiniHttpClient("127.0.0.1", self.port)
+            try:
+                while True:
+                    try:
+                        pod = q.get_nowait()
+                    except _queue.Empty:
+                        return
+                    try:
+                        self._schedule_one(conn, pod, record_latency)
+                    except Exception as exc:  # noqa: BLE001
+                        errors.append(exc)
+            finally:
+                conn.close()
 
-            await asyncio.gather(*[one(p) for p in pods])
+        threads = [threading.Thread(target=worker) for _ in range(n_workers)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        if errors:
+            raise errors[0]
+
+    def _schedule_one(self, conn, pod, record_latency: bool):
+        t0 = time.perf_counter()
+        for attempt in range(8):
+            status, body = conn.post_json(
+                "/scheduler/filter",
+                {"pod": pod, "nodenames": self.node_names})
+            ok = body.get("nodenames") or []
+            if not ok:
+                raise RuntimeError(f"no feasible node: {body}")
+            status, prio = conn.post_json(
+                "/scheduler/priorities", {"pod": pod, "nodenames": ok})
+            top = max(e["score"] for e in prio)
+            tied = [e["host"] for e in prio if e["score"] == top]
+            best = tied[(hash(pod["metadata"]["uid"]) + attempt) % len(tied)]
+            status, out = conn.post_json("/scheduler/bind", {
+                "podName": pod["metadata"]["name"],
+                "podNamespace": "default",
+                "podUID": pod["metadata"]["uid"],
+                "node": best})
+            if status == 200:
+                break
+            self.bind_retries += 1
+        else:
+            raise RuntimeError(f"bind kept failing: {out}")
+        if record_latency:
+            with self._lat_mu:
+                self.latencies.append(time.perf_counter() - t0)
 
     def _schedule_direct(self, pods, record_latency: bool):
         """--no-http: drive the handlers in-process (core profiling mode)."""
@@ -294,6 +366,8 @@ class BenchPipeline:
 
     def close(self):
         self.controller.stop()
+        if self._native is not None:
+            self._native.stop()
         if self._server is not None:
             self._server.should_exit = True
             self._server_thread.join(timeout=5)
@@ -316,6 +390,9 @@ def main():
     p.add_argument("--concurrency", type=int, default=16,
                    help="in-flight pods in the load generator")
     p.add_argument("--filter-threads", type=int, default=0)
+    p.add_argument("--server", default="native",
+                   choices=("native", "uvicorn"),
+                   help="HTTP front end: native C++ (default) or uvicorn")
     p.add_argument("--no-http", action="store_true",
                    help="bypass TCP; drive handlers in-process")
     p.add_argument("--no-verify", action="store_true",
@@ -402,6 +479,9 @@ def main():
                     "cards_per_node": args.cards,
                     "card": "MI355X 288GB",
                     "http": not args.no_http,
+                    "server": "none" if args.no_http else args.server,
+                    "native_stats": (pipe._native.stats()
+                                     if pipe._native else None),
                     "concurrency": args.concurrency,
                     "p50_filter_bind_ms": round(p50, 3) if p50 else None,
                     "p99_filter_bind_ms": round(p99, 3) if p99 else None,

@@ -15,6 +15,8 @@
 namespace py = pybind11;
 using namespace egs;
 
+void bind_native_server(py::module_& m);
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "MI355X-native elastic GPU scheduler core (C++)";
 
@@ -103,7 +105,9 @@ PYBIND11_MODULE(_core, m) {
       py::arg("devices"), py::arg("request"), py::arg("policy") = "binpack",
       py::arg("seed") = 0, py::arg("topology") = std::vector<std::vector<int>>{});
 
-  py::class_<ClusterState>(m, "ClusterState")
+  bind_native_server(m);
+
+  py::class_<ClusterState, std::shared_ptr<ClusterState>>(m, "ClusterState")
       .def(py::init<const std::string&, uint64_t, int>(), py::arg("policy") = "binpack",
            py::arg("seed") = 0, py::arg("threads") = 0)
       .def_property_readonly("policy", &ClusterState::policy)

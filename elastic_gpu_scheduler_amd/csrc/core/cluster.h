@@ -93,6 +93,10 @@ class ThreadPool {
   bool stop_ = false;
 };
 
+// Below this many nodes, the filter fan-out runs inline on the calling
+// thread (pool dispatch overhead > the per-node search cost).
+constexpr size_t kInlineFanout = 24;
+
 enum class AssumeVerdict : int {
   kOk = 0,
   kInfeasible = 1,
@@ -144,12 +148,19 @@ class ClusterState {
       }
     }
     std::vector<int> verdicts(names.size(), static_cast<int>(AssumeVerdict::kUnknownNode));
-    pool_.parallel_for(static_cast<int>(names.size()), [&](int i) {
+    auto task = [&](int i) {
       if (!allocs[i]) return;
       verdicts[i] = allocs[i]->assume(uid, req, *rater_)
                         ? static_cast<int>(AssumeVerdict::kOk)
                         : static_cast<int>(AssumeVerdict::kInfeasible);
-    });
+    };
+    // Small fan-outs run inline: per-node search is ~5-10 us, so pool
+    // dispatch overhead dominates below a few dozen nodes.
+    if (names.size() <= kInlineFanout) {
+      for (size_t i = 0; i < names.size(); ++i) task(static_cast<int>(i));
+    } else {
+      pool_.parallel_for(static_cast<int>(names.size()), task);
+    }
     return verdicts;
   }
 
@@ -164,10 +175,15 @@ class ClusterState {
       }
     }
     std::vector<double> scores(names.size(), kScoreMin);
-    pool_.parallel_for(static_cast<int>(names.size()), [&](int i) {
+    auto task = [&](int i) {
       if (!allocs[i]) return;
       scores[i] = allocs[i]->score(uid, req, *rater_);
-    });
+    };
+    if (names.size() <= kInlineFanout) {
+      for (size_t i = 0; i < names.size(); ++i) task(static_cast<int>(i));
+    } else {
+      pool_.parallel_for(static_cast<int>(names.size()), task);
+    }
     return scores;
   }
 

@@ -1,0 +1,116 @@
+// Python bindings for the native extender HTTP server.
+//
+// The server shares the scheduler's ClusterState: filter/priorities are
+// answered entirely in C++ (no GIL), everything else — bind (apiserver
+// writes), status, version, metrics, debug — is delegated to a Python
+// fallback callable `(method, path, body-bytes) -> (status, content_type,
+// body-bytes)`.
+#include <pybind11/functional.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <algorithm>
+#include <memory>
+
+#include "../httpd/extender.h"
+#include "../httpd/http_server.h"
+#include "cluster.h"
+
+namespace py = pybind11;
+using namespace egs;
+
+namespace {
+
+BareUnit parse_bare_unit(const std::string& s) {
+  if (s == "bytes") return BareUnit::Bytes;
+  if (s == "GiB") return BareUnit::GiB;
+  if (s == "MiB") return BareUnit::MiB;
+  return BareUnit::Auto;
+}
+
+class NativeExtenderServer {
+ public:
+  NativeExtenderServer(std::shared_ptr<ClusterState> state,
+                       const std::string& bare_unit, const std::string& host,
+                       int port, py::object fallback)
+      : core_(std::make_shared<ExtenderCore>(state, parse_bare_unit(bare_unit))),
+        fallback_(std::move(fallback)) {
+    auto core = core_;
+    py::object fb = fallback_;
+    egshttp::Handler handler = [core, fb](const egshttp::Request& req,
+                                          egshttp::Response* resp) {
+      if (req.method == "POST" && req.path == "/scheduler/filter") {
+        if (core->filter(req.body, &resp->body) == HandleStatus::Handled) return;
+      } else if (req.method == "POST" && req.path == "/scheduler/priorities") {
+        if (core->priorities(req.body, &resp->body) == HandleStatus::Handled)
+          return;
+      }
+      // Python fallback (bind, status, metrics, errors, cold nodes...).
+      core->counters.fallback.fetch_add(1, std::memory_order_relaxed);
+      py::gil_scoped_acquire gil;
+      try {
+        py::tuple out = fb(req.method, req.path, py::bytes(req.body));
+        resp->status = out[0].cast<int>();
+        resp->content_type = out[1].cast<std::string>();
+        resp->body = out[2].cast<std::string>();
+      } catch (const std::exception& e) {
+        core->counters.errors.fetch_add(1, std::memory_order_relaxed);
+        resp->status = 500;
+        resp->body = std::string("{\"error\": \"fallback failed: ") + e.what() +
+                     "\"}";
+      }
+    };
+    server_ = std::make_unique<egshttp::HttpServer>(host, port, handler);
+  }
+
+  ~NativeExtenderServer() { stop(); }
+
+  void start() { server_->start(); }
+  void stop() {
+    if (server_) server_->stop();
+  }
+  int port() const { return server_->port(); }
+
+  void note_filter(const std::string& uid) { core_->tracker.note(uid); }
+  double pop_filter_seconds(const std::string& uid) {
+    return core_->tracker.pop(uid);
+  }
+
+  py::dict stats() const {
+    py::dict d;
+    d["filter_native"] = core_->counters.filter_native.load();
+    d["priorities_native"] = core_->counters.priorities_native.load();
+    d["fallback"] = core_->counters.fallback.load();
+    d["errors"] = core_->counters.errors.load();
+    return d;
+  }
+
+ private:
+  std::shared_ptr<ExtenderCore> core_;
+  py::object fallback_;
+  std::unique_ptr<egshttp::HttpServer> server_;
+};
+
+}  // namespace
+
+void bind_native_server(py::module_& m) {
+  py::class_<NativeExtenderServer>(m, "NativeExtenderServer")
+      .def(py::init<std::shared_ptr<ClusterState>, const std::string&,
+                    const std::string&, int, py::object>(),
+           py::arg("state"), py::arg("bare_unit") = "auto",
+           py::arg("host") = "0.0.0.0", py::arg("port") = 0,
+           py::arg("fallback"))
+      .def("start", &NativeExtenderServer::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &NativeExtenderServer::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("port", &NativeExtenderServer::port)
+      .def("note_filter", &NativeExtenderServer::note_filter)
+      .def("pop_filter_seconds", &NativeExtenderServer::pop_filter_seconds)
+      .def("stats", &NativeExtenderServer::stats);
+
+  // JSON codec round-trip (exposed for tests of the native parser).
+  m.def("json_roundtrip", [](const std::string& s) {
+    return egsjson::dump(egsjson::parse(s));
+  });
+}

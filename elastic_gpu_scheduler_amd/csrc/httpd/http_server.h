@@ -1,0 +1,251 @@
+// Minimal HTTP/1.1 server for the native extender.
+//
+// The reference serves its extender webhook with Go's net/http + httprouter
+// (pkg/routes/routes.go); this is the MI355X rebuild's native equivalent: an
+// accept loop + thread-per-connection keep-alive server whose hot routes
+// (filter/priorities) never touch Python. Connection counts here are small
+// and long-lived (kube-scheduler holds a few keep-alive connections), so
+// thread-per-connection is the right simplicity/throughput tradeoff; a
+// handler callback decides each response (native fast path or a Python
+// fallback that acquires the GIL).
+#pragma once
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <cstring>
+#include <functional>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+namespace egshttp {
+
+struct Request {
+  std::string method;
+  std::string path;
+  std::string body;
+};
+
+struct Response {
+  int status = 200;
+  std::string content_type = "application/json";
+  std::string body;
+};
+
+using Handler = std::function<void(const Request&, Response*)>;
+
+inline const char* status_text(int code) {
+  switch (code) {
+    case 200: return "OK";
+    case 400: return "Bad Request";
+    case 404: return "Not Found";
+    case 500: return "Internal Server Error";
+    default: return "OK";
+  }
+}
+
+class HttpServer {
+ public:
+  HttpServer(const std::string& host, int port, Handler handler,
+             int max_connections = 512)
+      : host_(host), handler_(std::move(handler)),
+        max_connections_(max_connections) {
+    listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
+    int one = 1;
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_port = htons(static_cast<uint16_t>(port));
+    if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1)
+      addr.sin_addr.s_addr = INADDR_ANY;
+    if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)) < 0) {
+      ::close(listen_fd_);
+      throw std::runtime_error("bind() failed on port " + std::to_string(port));
+    }
+    socklen_t len = sizeof(addr);
+    getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&addr), &len);
+    port_ = ntohs(addr.sin_port);
+    if (::listen(listen_fd_, 128) < 0) {
+      ::close(listen_fd_);
+      throw std::runtime_error("listen() failed");
+    }
+  }
+
+  ~HttpServer() { stop(); }
+
+  int port() const { return port_; }
+
+  void start() {
+    running_ = true;
+    accept_thread_ = std::thread([this] { accept_loop(); });
+  }
+
+  void stop() {
+    if (!running_.exchange(false)) return;
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    if (accept_thread_.joinable()) accept_thread_.join();
+    // close live connections so their threads exit
+    {
+      std::lock_guard<std::mutex> g(conn_mu_);
+      for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+    }
+    for (auto& t : conn_threads_)
+      if (t.joinable()) t.join();
+    conn_threads_.clear();
+  }
+
+ private:
+  void accept_loop() {
+    while (running_) {
+      int fd = ::accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) {
+        if (!running_) return;
+        continue;
+      }
+      int one = 1;
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      {
+        std::lock_guard<std::mutex> g(conn_mu_);
+        if (static_cast<int>(conn_fds_.size()) >= max_connections_) {
+          ::close(fd);
+          continue;
+        }
+        conn_fds_.push_back(fd);
+        reap_finished_locked();
+        conn_threads_.emplace_back([this, fd] { connection_loop(fd); });
+      }
+    }
+  }
+
+  void reap_finished_locked() {
+    // bound the thread vector: join threads whose fds are gone
+    if (conn_threads_.size() < 1024) return;
+    for (auto& t : conn_threads_)
+      if (t.joinable()) t.detach();
+    conn_threads_.clear();
+  }
+
+  void connection_loop(int fd) {
+    std::string buf;
+    buf.reserve(8192);
+    char chunk[16384];
+    while (running_) {
+      // --- read one request ---
+      size_t header_end;
+      while ((header_end = buf.find("\r\n\r\n")) == std::string::npos) {
+        ssize_t n = ::recv(fd, chunk, sizeof(chunk), 0);
+        if (n <= 0) goto done;
+        buf.append(chunk, n);
+        if (buf.size() > (16u << 20)) goto done;  // 16 MiB header bound
+      }
+      {
+        Request req;
+        size_t line_end = buf.find("\r\n");
+        {
+          const std::string line = buf.substr(0, line_end);
+          size_t sp1 = line.find(' ');
+          size_t sp2 = line.find(' ', sp1 + 1);
+          if (sp1 == std::string::npos || sp2 == std::string::npos) goto done;
+          req.method = line.substr(0, sp1);
+          req.path = line.substr(sp1 + 1, sp2 - sp1 - 1);
+          size_t q = req.path.find('?');
+          if (q != std::string::npos) req.path.resize(q);
+        }
+        size_t content_length = 0;
+        bool keep_alive = true;
+        {
+          size_t pos = line_end + 2;
+          while (pos < header_end) {
+            size_t eol = buf.find("\r\n", pos);
+            std::string line = buf.substr(pos, eol - pos);
+            pos = eol + 2;
+            size_t colon = line.find(':');
+            if (colon == std::string::npos) continue;
+            std::string key = line.substr(0, colon);
+            for (auto& c : key) c = static_cast<char>(tolower(c));
+            size_t vstart = colon + 1;
+            while (vstart < line.size() && line[vstart] == ' ') ++vstart;
+            std::string value = line.substr(vstart);
+            if (key == "content-length") {
+              try {
+                content_length = std::stoul(value);
+              } catch (const std::exception&) {
+                goto done;
+              }
+              if (content_length > (64u << 20)) goto done;  // 64 MiB bound
+            } else if (key == "connection") {
+              for (auto& c : value) c = static_cast<char>(tolower(c));
+              keep_alive = value != "close";
+            }
+          }
+        }
+        size_t total = header_end + 4 + content_length;
+        while (buf.size() < total) {
+          ssize_t n = ::recv(fd, chunk, sizeof(chunk), 0);
+          if (n <= 0) goto done;
+          buf.append(chunk, n);
+        }
+        req.body = buf.substr(header_end + 4, content_length);
+        buf.erase(0, total);
+
+        // --- dispatch ---
+        Response resp;
+        try {
+          handler_(req, &resp);
+        } catch (const std::exception& e) {
+          resp.status = 500;
+          resp.body = std::string("{\"error\": \"") + e.what() + "\"}";
+        }
+
+        // --- write ---
+        std::string out;
+        out.reserve(resp.body.size() + 128);
+        out += "HTTP/1.1 ";
+        out += std::to_string(resp.status);
+        out += ' ';
+        out += status_text(resp.status);
+        out += "\r\ncontent-type: ";
+        out += resp.content_type;
+        out += "\r\ncontent-length: ";
+        out += std::to_string(resp.body.size());
+        out += keep_alive ? "\r\nconnection: keep-alive\r\n\r\n"
+                          : "\r\nconnection: close\r\n\r\n";
+        out += resp.body;
+        size_t off = 0;
+        while (off < out.size()) {
+          ssize_t n = ::send(fd, out.data() + off, out.size() - off, MSG_NOSIGNAL);
+          if (n <= 0) goto done;
+          off += n;
+        }
+        if (!keep_alive) goto done;
+      }
+    }
+  done:
+    ::close(fd);
+    std::lock_guard<std::mutex> g(conn_mu_);
+    conn_fds_.erase(std::remove(conn_fds_.begin(), conn_fds_.end(), fd),
+                    conn_fds_.end());
+  }
+
+  std::string host_;
+  Handler handler_;
+  int max_connections_;
+  int listen_fd_ = -1;
+  int port_ = 0;
+  std::atomic<bool> running_{false};
+  std::thread accept_thread_;
+  std::mutex conn_mu_;
+  std::vector<int> conn_fds_;
+  std::vector<std::thread> conn_threads_;
+};
+
+}  // namespace egshttp

@@ -39,6 +39,7 @@ log = logging.getLogger("egs.server")
 class ExtenderApp:
     def __init__(self, registry: SchedulerRegistry) -> None:
         self.registry = registry
+        self.native_server = None  # set by server.native.serve_native
         self._routes = {
             ("POST", "/scheduler/filter"): self.filter,
             ("POST", "/scheduler/priorities"): self.priorities,
@@ -50,26 +51,16 @@ class ExtenderApp:
             ("GET", "/debug/stacks"): self.debug_stacks,
         }
 
-    # ---- ASGI plumbing ---------------------------------------------------
+    # ---- shared sync dispatcher -----------------------------------------
 
-    async def __call__(self, scope, receive, send) -> None:
-        if scope["type"] != "http":
-            return
-        method = scope["method"]
-        path = scope["path"]
+    def handle(self, method: str, path: str, body: bytes):
+        """Dispatch one request; returns (status, content_type, body_bytes).
+        Used by the ASGI adapter AND as the native C++ server's Python
+        fallback (bind/status/metrics/cold-node requests)."""
         handler = self._routes.get((method, path))
         if handler is None:
-            await self._respond(send, 404, {"error": f"no route {method} {path}"})
-            return
-        body = b""
-        while True:
-            msg = await receive()
-            if msg["type"] == "http.request":
-                body += msg.get("body", b"")
-                if not msg.get("more_body"):
-                    break
-            else:
-                break
+            return (404, "application/json",
+                    json.dumps({"error": f"no route {method} {path}"}).encode())
         verb = path.rsplit("/", 1)[-1]
         with metrics.VERB_LATENCY.labels(verb).time():
             try:
@@ -82,14 +73,33 @@ class ExtenderApp:
                 log.exception("%s %s failed", method, path)
                 status, payload, raw = 500, {"error": f"{type(exc).__name__}: {exc}"}, None
                 metrics.REQUESTS.labels(verb, "exception").inc()
-        await self._respond(send, status, payload, raw)
+        ctype = "text/plain; charset=utf-8" if raw is not None and \
+            path in ("/debug/stacks",) else (
+                "text/plain; version=0.0.4; charset=utf-8"
+                if path == "/metrics" else "application/json")
+        return (status, ctype,
+                raw if raw is not None else json.dumps(payload).encode())
 
-    async def _respond(self, send, status: int, payload=None, raw: bytes = None,
-                       content_type: bytes = b"application/json") -> None:
-        body = raw if raw is not None else json.dumps(payload).encode()
+    # ---- ASGI plumbing ---------------------------------------------------
+
+    async def __call__(self, scope, receive, send) -> None:
+        if scope["type"] != "http":
+            return
+        method = scope["method"]
+        path = scope["path"]
+        body = b""
+        while True:
+            msg = await receive()
+            if msg["type"] == "http.request":
+                body += msg.get("body", b"")
+                if not msg.get("more_body"):
+                    break
+            else:
+                break
+        status, ctype, out = self.handle(method, path, body)
         await send({"type": "http.response.start", "status": status,
-                    "headers": [(b"content-type", content_type)]})
-        await send({"type": "http.response.body", "body": body})
+                    "headers": [(b"content-type", ctype.encode())]})
+        await send({"type": "http.response.body", "body": out})
 
     # ---- handlers --------------------------------------------------------
 
@@ -152,7 +162,12 @@ class ExtenderApp:
         except BindError as exc:
             return 500, {"error": str(exc)}, None
         metrics.PODS_SCHEDULED.inc()
-        metrics.TRACKER.saw_bind(obj.pod_uid(pod))
+        uid2 = obj.pod_uid(pod)
+        if metrics.TRACKER.saw_bind(uid2) is None and self.native_server is not None:
+            # the filter was answered by the C++ fast path; its tracker has t0
+            dt = self.native_server.pop_filter_seconds(uid2)
+            if dt >= 0:
+                metrics.FILTER_TO_BIND.observe(dt)
         return 200, {}, None
 
     def status(self, body: bytes):

@@ -1,0 +1,158 @@
+"""Native C++ HTTP server: wire parity with the Python ASGI app, fallback
+routing, keep-alive, and the JSON codec."""
+from __future__ import annotations
+
+import json
+import socket
+
+import httpx
+import pytest
+
+from elastic_gpu_scheduler_amd._native import core
+from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+from elastic_gpu_scheduler_amd.server.app import make_app
+from elastic_gpu_scheduler_amd.server.native import NativeFrontend
+from tests.conftest import make_node, make_pod
+
+GiB = 1024**3
+
+
+@pytest.fixture
+def native(fake_client):
+    fake_client.add_node(make_node("node-a"))
+    fake_client.add_node(make_node("node-b"))
+    registry = SchedulerRegistry(fake_client)
+    app = make_app(registry)
+    fe = NativeFrontend(app, host="127.0.0.1", port=0)
+    fe.start()
+    yield fake_client, registry, fe
+    fe.stop()
+
+
+def _client(fe):
+    return httpx.Client(base_url=f"http://127.0.0.1:{fe.port}", timeout=10.0)
+
+
+def test_json_codec_roundtrip():
+    cases = [
+        '{"a": 1, "b": [true, false, null], "c": {"d": "x\\ny"}}',
+        '{"nested": {"deep": {"deeper": [1.5, -2, 1e3]}}}',
+        '{"uni": "\\u00e9\\u4e2d\\ud83d\\ude00"}',
+        '[]', '{}', '"plain"', '42', '-3.25', 'true', 'null',
+    ]
+    for case in cases:
+        out = core.json_roundtrip(case)
+        assert json.loads(out) == json.loads(case), case
+    with pytest.raises(RuntimeError):
+        core.json_roundtrip("{broken")
+
+
+def test_native_filter_matches_python_app(native):
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("p", core=30, memory=64 * GiB))
+    # warm the node cache through the python path first
+    registry.default.assume(["node-a", "node-b"], pod)
+    with _client(fe) as c:
+        r = c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a", "node-b"]})
+        assert r.status_code == 200
+        assert sorted(r.json()["nodenames"]) == ["node-a", "node-b"]
+        stats = fe.stats()
+        assert stats["filter_native"] == 1
+
+        r = c.post("/scheduler/priorities",
+                   json={"pod": pod, "nodenames": ["node-a", "node-b"]})
+        scores = r.json()
+        assert {e["host"] for e in scores} == {"node-a", "node-b"}
+        assert all(isinstance(e["score"], int) for e in scores)
+        assert fe.stats()["priorities_native"] == 1
+
+
+def test_native_cold_node_falls_back_then_warms(native):
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("p", core=30, memory=64 * GiB))
+    with _client(fe) as c:
+        r = c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+        assert r.status_code == 200 and r.json()["nodenames"] == ["node-a"]
+        assert fe.stats()["fallback"] >= 1  # cold cache -> python
+        r = c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+        assert fe.stats()["filter_native"] == 1  # now native
+
+
+def test_native_full_schedule_cycle(native):
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("p", core=25, memory=48 * GiB))
+    with _client(fe) as c:
+        r = c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a", "node-b"]})
+        ok = r.json()["nodenames"]
+        r = c.post("/scheduler/priorities", json={"pod": pod, "nodenames": ok})
+        best = max(r.json(), key=lambda e: e["score"])["host"]
+        r = c.post("/scheduler/bind", json={
+            "podName": "p", "podNamespace": "default",
+            "podUID": pod["metadata"]["uid"], "node": best})
+        assert r.status_code == 200
+    bound = client.get_pod("default", "p")
+    assert bound["spec"]["nodeName"] == best
+    assert bound["metadata"]["annotations"]["elasticgpu.io/assumed"] == "true"
+
+
+def test_native_bad_json_gets_python_400(native):
+    _, _, fe = native
+    with _client(fe) as c:
+        r = c.post("/scheduler/filter", content=b"{nope",
+                   headers={"content-type": "application/json"})
+        assert r.status_code == 400
+
+
+def test_native_get_routes_fall_back(native):
+    client, registry, fe = native
+    with _client(fe) as c:
+        assert c.get("/healthz").json() == {"ok": True}
+        assert c.get("/version").json()["target"].startswith("MI355X")
+        assert b"egs_requests_total" in c.get("/metrics").content
+        assert c.get("/nope").status_code == 404
+
+
+def test_native_keep_alive_many_requests(native):
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("p", core=10, memory=GiB))
+    registry.default.assume(["node-a"], pod)
+    body = json.dumps({"pod": pod, "nodenames": ["node-a"]}).encode()
+    payload = (b"POST /scheduler/filter HTTP/1.1\r\nhost: t\r\n"
+               b"content-length: " + str(len(body)).encode() + b"\r\n\r\n" + body)
+    s = socket.create_connection(("127.0.0.1", fe.port))
+    try:
+        for _ in range(50):
+            s.sendall(payload)
+            data = b""
+            while b"\r\n\r\n" not in data:
+                data += s.recv(65536)
+            head, _, rest = data.partition(b"\r\n\r\n")
+            clen = int([l for l in head.split(b"\r\n")
+                        if l.lower().startswith(b"content-length")][0]
+                       .split(b":")[1])
+            while len(rest) < clen:
+                rest += s.recv(65536)
+            out = json.loads(rest[:clen])
+            assert out["nodenames"] == ["node-a"]
+    finally:
+        s.close()
+    assert fe.stats()["filter_native"] >= 50
+
+
+def test_native_filter_to_bind_latency_recorded(native):
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("p", core=25, memory=GiB))
+    registry.default._ensure_node("node-a")
+    with _client(fe) as c:
+        c.post("/scheduler/filter", json={"pod": pod, "nodenames": ["node-a"]})
+        assert fe.stats()["filter_native"] == 1
+        r = c.post("/scheduler/bind", json={
+            "podName": "p", "podNamespace": "default",
+            "podUID": pod["metadata"]["uid"], "node": "node-a"})
+        assert r.status_code == 200
+    # the native tracker entry was consumed by the bind handler
+    assert fe.server.pop_filter_seconds(pod["metadata"]["uid"]) < 0
