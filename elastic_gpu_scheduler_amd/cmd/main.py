@@ -47,6 +47,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--fake-cluster", type=int, default=0, metavar="NODES",
                    help="serve against an in-process fake apiserver with N "
                         "8x-MI355X nodes (local dev / benchmarking)")
+    p.add_argument("--server", default="native", choices=("native", "uvicorn"),
+                   help="HTTP front end: native C++ (default) or uvicorn ASGI")
     p.add_argument("--log-level", default="info")
     return p
 
@@ -113,20 +115,28 @@ def main(argv=None) -> int:
     signal.signal(signal.SIGINT, handle)
     signal.signal(signal.SIGTERM, handle)
 
-    import uvicorn
+    log.info("listening on %s:%d (policy=%s mode=%s server=%s)", args.host,
+             args.port, args.priority, args.mode, args.server)
+    if args.server == "native":
+        from elastic_gpu_scheduler_amd.server.native import NativeFrontend
 
-    config = uvicorn.Config(app, host=args.host, port=args.port,
-                            log_level=args.log_level, access_log=False)
-    server = uvicorn.Server(config)
-
-    def watch_stop():
+        fe = NativeFrontend(app, host=args.host, port=args.port)
+        fe.start()
         stop_event.wait()
-        server.should_exit = True
+        fe.stop()
+    else:
+        import uvicorn
 
-    threading.Thread(target=watch_stop, daemon=True).start()
-    log.info("listening on %s:%d (policy=%s mode=%s)", args.host, args.port,
-             args.priority, args.mode)
-    server.run()
+        config = uvicorn.Config(app, host=args.host, port=args.port,
+                                log_level=args.log_level, access_log=False)
+        server = uvicorn.Server(config)
+
+        def watch_stop():
+            stop_event.wait()
+            server.should_exit = True
+
+        threading.Thread(target=watch_stop, daemon=True).start()
+        server.run()
     controller.stop()
     return 0
 

@@ -251,14 +251,15 @@ class RealKubeClient(KubeClient):
     """
 
     def __init__(self, base_url: str, token: Optional[str] = None,
-                 verify: "bool | str" = True) -> None:
+                 verify: "bool | str" = True, transport=None) -> None:
         import httpx
 
         headers = {"Content-Type": "application/json"}
         if token:
             headers["Authorization"] = f"Bearer {token}"
         self._client = httpx.Client(base_url=base_url, headers=headers,
-                                    verify=verify, timeout=30.0)
+                                    verify=verify, timeout=30.0,
+                                    transport=transport)
         self._watch_stop = threading.Event()
 
     @classmethod
