@@ -48,9 +48,13 @@ inline bool parse_quantity_ll(const std::string& s, int64_t* out,
     ++i;
   }
   if (i == start) return false;
-  double num;
+  std::string num_s = s.substr(start, i - start);
+  bool is_int = num_s.find_first_of(".eE") == std::string::npos;
+  double num = 0.0;
+  int64_t num_i = 0;
   try {
-    num = std::stod(s.substr(start, i - start));
+    if (is_int) num_i = std::stoll(num_s);
+    else num = std::stod(num_s);
   } catch (const std::exception&) {
     return false;
   }
@@ -74,7 +78,13 @@ inline bool parse_quantity_ll(const std::string& s, int64_t* out,
   else if (suffix == "E") mult = 1e18;
   else if (suffix == "m") mult = 1e-3;
   else return false;
-  *out = static_cast<int64_t>(num * mult);
+  if (is_int) {
+    // exact integer scaling for integral multipliers (large byte counts)
+    if (mult >= 1.0) *out = num_i * static_cast<int64_t>(mult);
+    else *out = num_i / 1000;  // "m" milli
+  } else {
+    *out = static_cast<int64_t>(num * mult);
+  }
   return true;
 }
 

@@ -180,7 +180,16 @@ class ExtenderApp:
                                          self.registry.schedulers.values()})}, None
 
     def metrics(self, body: bytes):
-        return 200, None, metrics.render()
+        out = metrics.render()
+        if self.native_server is not None:
+            stats = self.native_server.stats()
+            lines = ["# HELP egs_native_requests_total requests answered by "
+                     "the C++ fast path / fallback",
+                     "# TYPE egs_native_requests_total counter"]
+            for key, val in stats.items():
+                lines.append(f'egs_native_requests_total{{kind="{key}"}} {val}')
+            out += ("\n".join(lines) + "\n").encode()
+        return 200, None, out
 
     def healthz(self, body: bytes):
         return 200, {"ok": True}, None

@@ -23,25 +23,33 @@ _DECIMAL = {"k": 10**3, "M": 10**6, "G": 10**9, "T": 10**12, "P": 10**15,
             "E": 10**18}
 
 _QTY_RE = re.compile(r"^\s*([+-]?[0-9.]+(?:[eE][+-]?[0-9]+)?)\s*([A-Za-z]*)\s*$")
+_INT_RE = re.compile(r"^[+-]?[0-9]+$")
 
 BARE_AUTO_GIB_THRESHOLD = 8192
 
 
 def parse_quantity(value: "str | int | float", bare_unit: str = "auto") -> int:
-    """Parse a k8s quantity into an integer (suffix-scaled, rounded down)."""
-    if isinstance(value, (int, float)):
-        num, suffix = float(value), ""
+    """Parse a k8s quantity into an integer (suffix-scaled, rounded down).
+
+    Integer numeric parts are scaled with exact integer arithmetic (large
+    byte counts like 4611686019G must not lose precision to float)."""
+    if isinstance(value, int):
+        num_s, suffix = str(value), ""
+    elif isinstance(value, float):
+        num_s, suffix = repr(value), ""
     else:
         m = _QTY_RE.match(str(value))
         if not m:
             raise ValueError(f"invalid quantity: {value!r}")
-        num, suffix = float(m.group(1)), m.group(2)
+        num_s, suffix = m.group(1), m.group(2)
+    is_int = _INT_RE.match(num_s) is not None
+    num = int(num_s) if is_int else float(num_s)
     if suffix in _BINARY:
         return int(num * _BINARY[suffix])
     if suffix in _DECIMAL:
         return int(num * _DECIMAL[suffix])
     if suffix == "m":
-        return int(num / 1000)
+        return int(num // 1000) if is_int else int(num / 1000)
     if suffix == "":
         return int(num)
     raise ValueError(f"unknown quantity suffix {suffix!r} in {value!r}")

@@ -156,3 +156,44 @@ def test_native_filter_to_bind_latency_recorded(native):
         assert r.status_code == 200
     # the native tracker entry was consumed by the bind handler
     assert fe.server.pop_filter_seconds(pod["metadata"]["uid"]) < 0
+
+
+def test_native_http_edge_cases(native):
+    """Connection: close honored; bad request line closes; metrics include
+    native counters."""
+    client, registry, fe = native
+    import socket as _socket
+
+    # Connection: close
+    s = _socket.create_connection(("127.0.0.1", fe.port))
+    s.sendall(b"GET /healthz HTTP/1.1\r\nhost: t\r\nconnection: close\r\n\r\n")
+    data = b""
+    while True:
+        chunk = s.recv(65536)
+        if not chunk:
+            break
+        data += chunk
+    assert b'{"ok": true}' in data
+    assert b"connection: close" in data.lower()
+    s.close()
+
+    # malformed request line -> server closes without crashing
+    s = _socket.create_connection(("127.0.0.1", fe.port))
+    s.sendall(b"BOGUS\r\n\r\n")
+    s.settimeout(5)
+    assert s.recv(1024) == b""  # closed
+    s.close()
+
+    # oversized content-length refused (connection closed)
+    s = _socket.create_connection(("127.0.0.1", fe.port))
+    s.sendall(b"POST /scheduler/filter HTTP/1.1\r\nhost: t\r\n"
+              b"content-length: 99999999999\r\n\r\n")
+    s.settimeout(5)
+    assert s.recv(1024) == b""
+    s.close()
+
+    # server still serves after the bad clients
+    with _client(fe) as c:
+        assert c.get("/healthz").status_code == 200
+        m = c.get("/metrics").content
+        assert b"egs_native_requests_total" in m
