@@ -95,7 +95,7 @@ class ThreadPool {
 
 // Below this many nodes, the filter fan-out runs inline on the calling
 // thread (pool dispatch overhead > the per-node search cost).
-constexpr size_t kInlineFanout = 24;
+constexpr size_t kInlineFanout = 64;
 
 enum class AssumeVerdict : int {
   kOk = 0,
