@@ -144,6 +144,45 @@ def _via_torch() -> List[Dict[str, Any]]:
     return cards
 
 
+def parse_compute_partition(payload: str) -> Dict[int, str]:
+    """Parse `rocm-smi --showcomputepartition --json`:
+    {"card0": {"Compute Partition": "SPX"}, ...} -> {0: "SPX"}.
+
+    MI355X partition modes change what a "card" is: SPX exposes one 288 GB
+    device per OAM; CPX exposes 8 partitions of ~36 GB each. The scheduler
+    treats whatever is enumerated as the card vector, and the partition tag
+    published with the inventory lets operators see which mode produced it.
+    """
+    try:
+        data = json.loads(payload)
+    except json.JSONDecodeError:
+        return {}
+    out = {}
+    for key, entry in data.items():
+        if not key.startswith("card") or not isinstance(entry, dict):
+            continue
+        for field in ("Compute Partition", "compute_partition"):
+            if field in entry:
+                try:
+                    out[int(key[len("card"):])] = str(entry[field]).strip()
+                except ValueError:
+                    pass
+                break
+    return out
+
+
+def _query_partitions() -> Dict[int, str]:
+    exe = shutil.which("rocm-smi")
+    if not exe:
+        return {}
+    try:
+        out = subprocess.run([exe, "--showcomputepartition", "--json"],
+                             capture_output=True, text=True, timeout=30)
+        return parse_compute_partition(out.stdout)
+    except (subprocess.SubprocessError, OSError):
+        return {}
+
+
 def discover(prefer: str = "auto") -> List[Dict[str, Any]]:
     """Discover the node's GPU cards. `prefer` forces one source in tests."""
     sources = {
@@ -161,5 +200,9 @@ def discover(prefer: str = "auto") -> List[Dict[str, Any]]:
             log.debug("inventory source %s failed", name, exc_info=True)
             cards = []
         if cards:
+            partitions = _query_partitions()
+            for c in cards:
+                if c["index"] in partitions:
+                    c["partition"] = partitions[c["index"]]
             return cards
     return []

@@ -15,6 +15,7 @@
 //     scheduler.go:44).
 #pragma once
 
+#include <algorithm>
 #include <chrono>
 #include <cstdint>
 #include <mutex>
@@ -208,6 +209,10 @@ class NodeAllocator {
   }
 
   void gc_assumed_locked() {
+    // A pod assumes on EVERY filtered node but binds on one, so entries for
+    // the losing nodes linger. Sweep expired entries once the map is large,
+    // and hard-cap the map by evicting the oldest half — sustained
+    // scheduling must not grow memory without bound.
     if (assumed_.size() < 1024) return;  // amortise: only sweep when large
     auto cutoff = now() - kAssumeTTL;
     for (auto it = assumed_.begin(); it != assumed_.end();) {
@@ -216,6 +221,15 @@ class NodeAllocator {
       else
         ++it;
     }
+    constexpr size_t kHardCap = 8192;
+    if (assumed_.size() <= kHardCap) return;
+    std::vector<std::pair<Clock::time_point, std::string>> order;
+    order.reserve(assumed_.size());
+    for (const auto& [uid, e] : assumed_) order.emplace_back(e.at, uid);
+    std::nth_element(order.begin(), order.begin() + order.size() / 2,
+                     order.end());
+    for (size_t i = 0; i < order.size() / 2; ++i)
+      assumed_.erase(order[i].second);
   }
 
   std::string name_;

@@ -112,3 +112,11 @@ def test_discover_empty_on_cpu_only_box():
         return  # running on a GPU box: covered by gpu-marked tests
     assert inv.discover() == []
     assert topo.discover(8) == topo.default_hive(8)
+
+
+def test_parse_compute_partition():
+    payload = json.dumps({"card0": {"Compute Partition": "SPX"},
+                          "card1": {"Compute Partition": "CPX"},
+                          "system": {"driver": "x"}})
+    assert inv.parse_compute_partition(payload) == {0: "SPX", 1: "CPX"}
+    assert inv.parse_compute_partition("junk") == {}
