@@ -178,6 +178,35 @@ def config4():
     return {"config": "4: 64 mixed pods, binpack vs spread", **out}
 
 
+def packing_quality():
+    """BASELINE's third metric: binpack packing quality. Schedule a fixed
+    mixed workload (32 fractional pods) under each policy and report how
+    many cards remain FULLY free — free whole cards are the currency of
+    packing quality (they can still take whole-card pods)."""
+    out = {}
+    for policy in ("binpack", "spread", "random"):
+        client = make_client([("node-0", 8, None)])
+        h = Harness(client, policy=policy)
+        try:
+            for i in range(32):
+                req = {"elasticgpu.io/gpu-core": [10, 25, 15][i % 3],
+                       "elasticgpu.io/gpu-memory": [16, 48, 32][i % 3] * GiB}
+                pod = make_pod(client, f"q{i}", req)
+                h.schedule(pod, ["node-0"])
+            devs = h.registry.default.state.node_devices("node-0")
+            whole_free = sum(1 for d in devs
+                             if d.core_avail == d.core_total and
+                             d.mem_avail == d.mem_total)
+            used = [round(1 - d.core_avail / d.core_total, 2) for d in devs]
+            out[policy] = {"whole_free_cards": whole_free,
+                           "core_utilization": used}
+        finally:
+            h.close()
+    assert out["binpack"]["whole_free_cards"] >=         out["spread"]["whole_free_cards"], out
+    return {"config": "packing quality: 32 mixed fractional pods, 8 cards",
+            **out}
+
+
 def config5():
     """gpu-core=400 on a partitioned-hive 8-card node: must stay in-hive."""
     hops = [[0 if i == j else (1 if (i < 4) == (j < 4) else 3)
@@ -210,7 +239,8 @@ def main():
     p.add_argument("--json", default="")
     args = p.parse_args()
     results = []
-    for fn in (config1, config2, config3, config4, config5):
+    for fn in (config1, config2, config3, config4, config5,
+               packing_quality):
         t0 = time.time()
         r = fn()
         r["wall_s"] = round(time.time() - t0, 2)

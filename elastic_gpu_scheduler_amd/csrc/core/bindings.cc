@@ -146,6 +146,13 @@ PYBIND11_MODULE(_core, m) {
              }
              return snap;
            })
+      .def("node_assumed_count",
+           [](ClusterState& cs, const std::string& name) {
+             auto alloc = cs.get(name);
+             if (!alloc) throw std::runtime_error("unknown node " + name);
+             py::gil_scoped_release rel;
+             return alloc->assumed_count();
+           })
       .def("node_pods", [](ClusterState& cs, const std::string& name) {
         auto alloc = cs.get(name);
         if (!alloc) throw std::runtime_error("unknown node " + name);

@@ -128,6 +128,11 @@ class NodeAllocator {
     return devices_;
   }
 
+  int assumed_count() {
+    std::lock_guard<std::mutex> g(mu_);
+    return static_cast<int>(assumed_.size());
+  }
+
   std::vector<std::string> pod_uids() {
     std::lock_guard<std::mutex> g(mu_);
     std::vector<std::string> out;

@@ -136,3 +136,27 @@ def test_forget_while_assume_storm():
     # nothing left allocated
     devs = c.node_devices("n")
     assert all(d.core_avail == 100 and d.mem_avail == 288 * GiB for d in devs)
+
+
+def test_soak_assumed_cache_stays_bounded():
+    """Sustained schedule/release churn must not grow the per-node assume
+    cache without bound (each pod assumes on every node but binds on one)."""
+    c = core.ClusterState("binpack", 0, 0)
+    nodes = [f"n{i}" for i in range(4)]
+    for n in nodes:
+        c.add_node(n, [core.Device(100, 100, 288 * GiB, 288 * GiB)
+                       for _ in range(8)], [])
+    req = [core.GPUUnit(0, 20, 8 * GiB)]
+    for wave in range(60):
+        uids = [f"w{wave}-p{i}" for i in range(100)]
+        for uid in uids:
+            c.assume(nodes, uid, req)
+        for uid in uids:
+            c.allocate(nodes[hash(uid) % 4], uid, req)
+        for uid in uids:
+            c.forget_pod(uid)
+    for n in nodes:
+        # TTL is 300s so nothing expires inside the test; the hard cap must
+        # hold the line instead (8192 per node).
+        assert c.node_assumed_count(n) <= 8192
+        assert c.node_pods(n) == []
