@@ -49,6 +49,12 @@ def build_parser() -> argparse.ArgumentParser:
                         "8x-MI355X nodes (local dev / benchmarking)")
     p.add_argument("--server", default="native", choices=("native", "uvicorn"),
                    help="HTTP front end: native C++ (default) or uvicorn ASGI")
+    p.add_argument("--leader-elect", action="store_true",
+                   help="Lease-based leader election: standby replicas wait "
+                        "for the lease instead of double-booking GPUs (the "
+                        "reference supports only replicas=1)")
+    p.add_argument("--leader-identity",
+                   default=os.environ.get("POD_NAME", "") or os.uname().nodename)
     p.add_argument("--log-level", default="info")
     return p
 
@@ -88,6 +94,26 @@ def main(argv=None) -> int:
         if args.kubeconf:
             os.environ["KUBECONFIG"] = args.kubeconf
         client = RealKubeClient.from_env()
+
+    if args.leader_elect:
+        from elastic_gpu_scheduler_amd.k8s.leader import LeaderElector
+
+        elector = LeaderElector(client, "elastic-gpu-scheduler-amd",
+                                args.leader_identity)
+        acquired = threading.Event()
+        lost = threading.Event()
+        threading.Thread(target=elector.run,
+                         args=(acquired.set, lost.set), daemon=True).start()
+        log.info("waiting for leadership (%s)...", args.leader_identity)
+        acquired.wait()
+        log.info("leadership acquired")
+
+        def watch_loss():
+            lost.wait()
+            log.error("leadership lost; exiting for restart")
+            os._exit(1)
+
+        threading.Thread(target=watch_loss, daemon=True).start()
 
     from elastic_gpu_scheduler_amd.controller.controller import Controller
     from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry

@@ -69,6 +69,16 @@ class KubeClient:
     def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
         raise NotImplementedError
 
+    # leases (leader election)
+    def get_lease(self, namespace: str, name: str) -> Dict[str, Any]:
+        raise NotImplementedError
+
+    def create_lease(self, namespace: str, lease: Dict[str, Any]) -> Dict[str, Any]:
+        raise NotImplementedError
+
+    def update_lease(self, namespace: str, lease: Dict[str, Any]) -> Dict[str, Any]:
+        raise NotImplementedError
+
     # watch
     def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
         """Register a pod watch; returns an unsubscribe callable."""
@@ -90,6 +100,7 @@ class FakeKubeClient(KubeClient):
         self._pods: Dict[str, Pod] = {}   # "ns/name" -> pod
         self._nodes: Dict[str, Node] = {}
         self._events: List[Dict[str, Any]] = []
+        self._leases: Dict[str, Dict[str, Any]] = {}
         self._rv = 0
         self._watchers: List[WatchHandler] = []
 
@@ -224,6 +235,37 @@ class FakeKubeClient(KubeClient):
         with self._mu:
             self._events.append(_jcopy(event))
 
+    def get_lease(self, namespace: str, name: str) -> Dict[str, Any]:
+        with self._mu:
+            key = self._key(namespace, name)
+            if key not in self._leases:
+                raise NotFoundError(f"lease {key} not found")
+            return _jcopy(self._leases[key])
+
+    def create_lease(self, namespace: str, lease: Dict[str, Any]) -> Dict[str, Any]:
+        with self._mu:
+            key = self._key(namespace, lease["metadata"]["name"])
+            if key in self._leases:
+                raise ConflictError(f"lease {key} exists")
+            lease = _jcopy(lease)
+            lease["metadata"]["resourceVersion"] = self._next_rv()
+            self._leases[key] = lease
+            return _jcopy(lease)
+
+    def update_lease(self, namespace: str, lease: Dict[str, Any]) -> Dict[str, Any]:
+        with self._mu:
+            key = self._key(namespace, lease["metadata"]["name"])
+            if key not in self._leases:
+                raise NotFoundError(f"lease {key} not found")
+            current = self._leases[key]
+            if lease["metadata"].get("resourceVersion") != \
+                    current["metadata"]["resourceVersion"]:
+                raise ConflictError(f"lease {key}: resourceVersion mismatch")
+            lease = _jcopy(lease)
+            lease["metadata"]["resourceVersion"] = self._next_rv()
+            self._leases[key] = lease
+            return _jcopy(lease)
+
     @property
     def events(self) -> List[Dict[str, Any]]:
         with self._mu:
@@ -345,6 +387,21 @@ class RealKubeClient(KubeClient):
     def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
         self._check(self._client.post(
             f"/api/v1/namespaces/{namespace}/events", content=json.dumps(event)))
+
+    def get_lease(self, namespace: str, name: str) -> Pod:
+        return self._check(self._client.get(
+            f"/apis/coordination.k8s.io/v1/namespaces/{namespace}/leases/{name}"))
+
+    def create_lease(self, namespace: str, lease: Pod) -> Pod:
+        return self._check(self._client.post(
+            f"/apis/coordination.k8s.io/v1/namespaces/{namespace}/leases",
+            content=json.dumps(lease)))
+
+    def update_lease(self, namespace: str, lease: Pod) -> Pod:
+        name = lease["metadata"]["name"]
+        return self._check(self._client.put(
+            f"/apis/coordination.k8s.io/v1/namespaces/{namespace}/leases/{name}",
+            content=json.dumps(lease)))
 
     def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
         stop = threading.Event()
