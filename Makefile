@@ -21,3 +21,11 @@ image:
 clean:
 	rm -f elastic_gpu_scheduler_amd/*.so
 	find . -name __pycache__ -type d -exec rm -rf {} +
+
+# Race-detection harness: the allocator core under ThreadSanitizer
+# (the reference ships no race detection at all — SURVEY.md §5).
+tsan-stress:
+	g++ -O1 -g -std=c++17 -fsanitize=thread -pthread \
+	  -Ielastic_gpu_scheduler_amd/csrc/core \
+	  elastic_gpu_scheduler_amd/csrc/stress/stress_main.cc \
+	  -o build_tsan_stress && ./build_tsan_stress

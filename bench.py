@@ -5,9 +5,10 @@ extender.
 Measures the BASELINE.json headline metric — pods scheduled/sec and p50
 filter->bind latency — on BASELINE config #4's workload: batches of 64 mixed
 gpu-core/gpu-memory pods scheduled onto 8x-MI355X-288GB nodes, binpack
-policy, through the FULL pipeline: real HTTP (uvicorn TCP) -> extender
-filter -> priorities -> bind (annotation write + binding) -> controller
-release, backed by an in-process fake apiserver (there is no cluster on the
+policy, through the FULL pipeline: real HTTP over TCP (native C++ server by
+default; --server uvicorn for the ASGI stack) -> extender filter ->
+priorities -> bind (annotation write + binding) -> controller release,
+backed by an in-process fake apiserver (there is no cluster on the
 bench box; the reference's apiserver round-trips are replaced by the fake's
 in-memory writes for BOTH warm and timed phases, stated here so the number
 is interpretable).

@@ -1,0 +1,72 @@
+// Standalone allocator stress binary for ThreadSanitizer runs.
+//
+// The reference configures no race detection at all (no -race anywhere,
+// SURVEY.md §5); this binary is the rebuild's race-detection harness: it
+// hammers ClusterState's verbs from many threads and is built with
+// -fsanitize=thread by `make tsan-stress` (and exercised by
+// tests/test_tsan_stress.py). Exit 0 = clean under TSAN.
+#include <atomic>
+#include <cstdio>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "../core/cluster.h"
+
+using namespace egs;
+
+int main() {
+  constexpr int64_t GiB = 1024LL * 1024 * 1024;
+  ClusterState cs("binpack", 0, 4);
+  const int kNodes = 4;
+  for (int i = 0; i < kNodes; ++i) {
+    std::vector<Device> devs(8);
+    for (auto& d : devs) {
+      d.mem_total = d.mem_avail = 288 * GiB;
+    }
+    cs.add_node("n" + std::to_string(i), devs, {});
+  }
+  std::vector<std::string> names;
+  for (int i = 0; i < kNodes; ++i) names.push_back("n" + std::to_string(i));
+
+  std::atomic<int> errors{0};
+  auto worker = [&](int wid) {
+    GPURequest req{GPUUnit{0, 10 + (wid % 5) * 10, 4 * GiB}};
+    for (int it = 0; it < 200; ++it) {
+      std::string uid = "w" + std::to_string(wid) + "-" + std::to_string(it);
+      cs.assume(names, uid, req);
+      cs.score(names, uid, req);
+      try {
+        cs.allocate(names[(wid + it) % kNodes], uid, req);
+      } catch (const std::exception&) {
+        continue;  // node full: fine
+      }
+      if (it % 3 != 0) cs.forget_pod(uid);
+    }
+  };
+  std::vector<std::thread> threads;
+  for (int w = 0; w < 16; ++w) threads.emplace_back(worker, w);
+  // concurrent node churn: add/remove an extra node while scheduling runs
+  threads.emplace_back([&] {
+    for (int i = 0; i < 50; ++i) {
+      std::vector<Device> devs(8);
+      for (auto& d : devs) d.mem_total = d.mem_avail = 288 * GiB;
+      cs.add_node("hot", devs, {});
+      cs.remove_node("hot");
+    }
+  });
+  for (auto& t : threads) t.join();
+
+  // availability must never be negative
+  for (const auto& n : names) {
+    auto alloc = cs.get(n);
+    for (const auto& d : alloc->snapshot()) {
+      if (d.core_avail < 0 || d.mem_avail < 0) {
+        std::fprintf(stderr, "negative availability on %s\n", n.c_str());
+        return 1;
+      }
+    }
+  }
+  std::printf("stress ok\n");
+  return errors.load() ? 1 : 0;
+}

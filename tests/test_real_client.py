@@ -116,3 +116,48 @@ def test_patch_node_and_events(api):
     assert mock.patches[0][1] == {"metadata": {"annotations": {"a": "b"}}}
     client.create_event("default", {"reason": "Scheduled"})
     assert mock.events[0]["reason"] == "Scheduled"
+
+
+def test_scheduler_service_over_rest_client(api):
+    """Full verb cycle through RealKubeClient (REST) instead of the fake:
+    the interface seam the reference lacks."""
+    from elastic_gpu_scheduler_amd.scheduler.service import GPUUnitScheduler
+
+    mock, client = api
+    # apiserver knows one 8-card MI355X node
+    GiB = 1024**3
+
+    real_get_node = mock.handler
+
+    def handler(request):
+        if request.method == "GET" and request.url.path == "/api/v1/nodes/gpu-1":
+            return httpx.Response(200, json={
+                "metadata": {"name": "gpu-1"},
+                "status": {"allocatable": {
+                    "elasticgpu.io/gpu-core": "800",
+                    "elasticgpu.io/gpu-memory": str(8 * 288 * GiB)}}})
+        return real_get_node(request)
+
+    client._client._transport = httpx.MockTransport(handler)
+
+    mock.pods["default/p"] = {
+        "metadata": {"name": "p", "namespace": "default", "uid": "u-1",
+                     "resourceVersion": "1"},
+        "spec": {"containers": [{"name": "c", "resources": {"requests": {
+            "elasticgpu.io/gpu-core": "30",
+            "elasticgpu.io/gpu-memory": str(48 * GiB)}}}]},
+        "status": {"phase": "Pending"},
+    }
+    sch = GPUUnitScheduler(client)
+    pod = client.get_pod("default", "p")
+    ok, failed = sch.assume(["gpu-1", "ghost"], pod)
+    assert ok == ["gpu-1"], failed
+    scores = sch.score(["gpu-1"], pod)
+    assert 0 <= scores[0] <= 10
+    sch.bind("gpu-1", pod)
+    # the annotation Update + Binding both hit the REST surface
+    assert mock.bindings[0]["target"]["name"] == "gpu-1"
+    stored = mock.pods["default/p"]
+    assert stored["metadata"]["annotations"]["elasticgpu.io/container-c"] \
+        in {str(i) for i in range(8)}
+    assert mock.events, "bind must emit a scheduling event"

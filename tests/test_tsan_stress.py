@@ -1,0 +1,25 @@
+"""Race detection: build the native-core stress harness with
+ThreadSanitizer and run it (SURVEY.md §5 — the reference has no race
+detection; this is the rebuild's)."""
+from __future__ import annotations
+
+import subprocess
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def test_core_clean_under_tsan(tmp_path):
+    binary = tmp_path / "stress"
+    build = subprocess.run(
+        ["g++", "-O1", "-g", "-std=c++17", "-fsanitize=thread", "-pthread",
+         "-I", str(REPO / "elastic_gpu_scheduler_amd/csrc/core"),
+         str(REPO / "elastic_gpu_scheduler_amd/csrc/stress/stress_main.cc"),
+         "-o", str(binary)],
+        capture_output=True, text=True, timeout=300)
+    assert build.returncode == 0, build.stderr[-3000:]
+    run = subprocess.run([str(binary)], capture_output=True, text=True,
+                         timeout=300,
+                         env={"TSAN_OPTIONS": "halt_on_error=1"})
+    assert run.returncode == 0, (run.stdout[-1000:], run.stderr[-3000:])
+    assert "stress ok" in run.stdout
