@@ -21,6 +21,7 @@ from __future__ import annotations
 
 import json
 import logging
+import queue
 import threading
 import time
 from typing import Any, Dict, List, Optional
@@ -56,6 +57,12 @@ class GPUUnitScheduler:
         # fast-path membership cache over state's node map (adds only; a
         # removed node falls back to the authoritative native check)
         self._known_nodes: set = set()
+        # events are emitted asynchronously: an apiserver event write must
+        # never sit on the bind critical path
+        self._event_q: "queue.Queue" = queue.Queue(maxsize=4096)
+        self._event_thread = threading.Thread(target=self._event_loop,
+                                              name="egs-events", daemon=True)
+        self._event_thread.start()
         self.warm_start()
 
     # ---- cache management ------------------------------------------------
@@ -256,7 +263,29 @@ class GPUUnitScheduler:
 
     def _emit_event(self, pod: Dict[str, Any], reason: str, message: str) -> None:
         try:
-            self.client.create_event(obj.pod_namespace(pod), {
+            self._event_q.put_nowait((obj.pod_namespace(pod), pod, reason,
+                                      message))
+        except queue.Full:
+            log.debug("event queue full; dropping %s event", reason)
+
+    def _event_loop(self) -> None:
+        while True:
+            ns, pod, reason, message = self._event_q.get()
+            try:
+                self._write_event(ns, pod, reason, message)
+            except Exception:
+                log.debug("event emit failed", exc_info=True)
+
+    def flush_events(self, timeout: float = 5.0) -> None:
+        """Wait for queued events to drain (tests)."""
+        deadline = time.time() + timeout
+        while not self._event_q.empty() and time.time() < deadline:
+            time.sleep(0.01)
+
+    def _write_event(self, ns: str, pod: Dict[str, Any], reason: str,
+                     message: str) -> None:
+        try:
+            self.client.create_event(ns, {
                 "metadata": {"generateName": "egs-"},
                 "involvedObject": {
                     "kind": "Pod",

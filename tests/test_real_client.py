@@ -160,4 +160,5 @@ def test_scheduler_service_over_rest_client(api):
     stored = mock.pods["default/p"]
     assert stored["metadata"]["annotations"]["elasticgpu.io/container-c"] \
         in {str(i) for i in range(8)}
+    sch.flush_events()
     assert mock.events, "bind must emit a scheduling event"
