@@ -217,8 +217,14 @@ class NodeAllocator {
     // A pod assumes on EVERY filtered node but binds on one, so entries for
     // the losing nodes linger. Sweep expired entries once the map is large,
     // and hard-cap the map by evicting the oldest half — sustained
-    // scheduling must not grow memory without bound.
+    // scheduling must not grow memory without bound. The sweep itself is
+    // O(map): gate it by time so steady-state assumes stay O(1) (an
+    // every-call sweep measurably throttled sustained throughput).
     if (assumed_.size() < 1024) return;  // amortise: only sweep when large
+    auto t = now();
+    if (assumed_.size() < 8192 && t - last_gc_ < std::chrono::seconds(5))
+      return;
+    last_gc_ = t;
     auto cutoff = now() - kAssumeTTL;
     for (auto it = assumed_.begin(); it != assumed_.end();) {
       if (it->second.at < cutoff)
@@ -237,6 +243,7 @@ class NodeAllocator {
       assumed_.erase(order[i].second);
   }
 
+  Clock::time_point last_gc_ = Clock::time_point::min();
   std::string name_;
   std::vector<Device> devices_;
   Topology topo_;

@@ -227,11 +227,13 @@ class GPUUnitScheduler:
     def forget_pod(self, pod: Dict[str, Any]) -> None:
         uid = obj.pod_uid(pod)
         with self._released_mu:
+            # dict preserves insertion order: evicting the oldest entry keeps
+            # the tombstone set bounded in O(1) per forget (a wholesale
+            # rebuild here measurably decayed sustained throughput)
+            self._released.pop(uid, None)
             self._released[uid] = time.time()
-            if len(self._released) > 4096:  # bounded tombstone set
-                cutoff = time.time() - 600
-                self._released = {k: v for k, v in self._released.items()
-                                  if v > cutoff}
+            while len(self._released) > 4096:
+                self._released.pop(next(iter(self._released)))
         self.state.forget_pod(uid)
 
     def known_pod(self, pod: Dict[str, Any]) -> bool:
