@@ -49,7 +49,6 @@ class GPUUnitScheduler:
         self.policy = policy
         self.bare_unit = bare_unit
         self.state = core.ClusterState(policy, seed, threads)
-        self._metrics_cbs: List[Any] = []
         # released-pod tombstones (reference releasedPodMap, scheduler.go:47):
         # a DELETE seen before the final MODIFIED must not re-add the pod.
         self._released: Dict[str, float] = {}
