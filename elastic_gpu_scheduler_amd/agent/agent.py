@@ -73,10 +73,18 @@ class NodeAgent:
         }
 
     def publish(self) -> Dict[str, str]:
+        """Publish inventory/topology annotations AND the elasticgpu.io
+        allocatable quantities onto the Node (the role the reference
+        delegates to its device plugin's node updates)."""
         if self.client is None:
             raise RuntimeError("NodeAgent.publish needs a KubeClient")
         ann = self.annotations()
         self.client.patch_node_annotations(self.node_name, ann)
+        try:
+            self.client.patch_node_allocatable(self.node_name,
+                                               self.allocatable())
+        except NotImplementedError:
+            pass  # client without status-subresource support
         return ann
 
     def node_object(self) -> Dict[str, Any]:

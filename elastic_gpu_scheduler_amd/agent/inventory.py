@@ -144,6 +144,65 @@ def _via_torch() -> List[Dict[str, Any]]:
     return cards
 
 
+def parse_amd_smi_static(payload: str) -> List[Dict[str, Any]]:
+    """Parse `amd-smi static --json` into cards.
+
+    amd-smi (the supported successor of rocm-smi) emits a list of per-GPU
+    objects: [{"gpu": 0, "vram": {"size": {"value": 294912, "unit": "MB"}},
+    "asic": {"market_name": ...}}, ...]. Field layouts drifted across ROCm
+    releases; this accepts the common shapes.
+    """
+    try:
+        data = json.loads(payload)
+    except json.JSONDecodeError:
+        return []
+    if isinstance(data, dict):
+        data = data.get("gpu_data", data.get("gpus", []))
+    if not isinstance(data, list):
+        return []
+    cards = []
+    for entry in data:
+        if not isinstance(entry, dict):
+            continue
+        idx = entry.get("gpu", entry.get("gpu_id", len(cards)))
+        vram = entry.get("vram", {})
+        size = vram.get("size", vram.get("vram_size", {}))
+        mem_bytes = None
+        if isinstance(size, dict):
+            val = size.get("value")
+            unit = str(size.get("unit", "MB")).upper()
+            if val is not None:
+                mult = {"B": 1, "KB": 1024, "MB": 1024**2,
+                        "GB": 1024**3}.get(unit, 1024**2)
+                mem_bytes = int(float(val) * mult)
+        elif isinstance(size, (int, float)):
+            mem_bytes = int(size) * 1024**2
+        if mem_bytes is None:
+            continue
+        asic = entry.get("asic", {}) if isinstance(entry.get("asic"), dict) else {}
+        cards.append({
+            "index": int(idx),
+            "name": asic.get("market_name", "AMD GPU"),
+            "gcn_arch": asic.get("target_graphics_version", ""),
+            "memory_bytes": mem_bytes,
+            "core": t.GPU_CORE_EACH_CARD,
+            "source": "amd-smi",
+        })
+    return cards
+
+
+def _via_amd_smi() -> List[Dict[str, Any]]:
+    exe = shutil.which("amd-smi")
+    if not exe:
+        return []
+    try:
+        out = subprocess.run([exe, "static", "--json"], capture_output=True,
+                             text=True, timeout=30)
+        return parse_amd_smi_static(out.stdout)
+    except (subprocess.SubprocessError, OSError):
+        return []
+
+
 def parse_compute_partition(payload: str) -> Dict[int, str]:
     """Parse `rocm-smi --showcomputepartition --json`:
     {"card0": {"Compute Partition": "SPX"}, ...} -> {0: "SPX"}.
@@ -188,12 +247,13 @@ def discover(prefer: str = "auto") -> List[Dict[str, Any]]:
     sources = {
         "gpuprobe": _via_gpuprobe,
         "amdsmi": _via_amdsmi,
+        "amd-smi": _via_amd_smi,
         "rocm-smi": _via_rocm_smi,
         "torch": _via_torch,
     }
     if prefer != "auto":
         return sources[prefer]()
-    for name in ("gpuprobe", "amdsmi", "rocm-smi", "torch"):
+    for name in ("gpuprobe", "amdsmi", "amd-smi", "rocm-smi", "torch"):
         try:
             cards = sources[name]()
         except Exception:

@@ -64,6 +64,12 @@ class KubeClient:
     def patch_node_annotations(self, name: str, annotations: Dict[str, str]) -> Node:
         raise NotImplementedError
 
+    def patch_node_allocatable(self, name: str,
+                               allocatable: Dict[str, str]) -> Node:
+        """Merge extended-resource quantities into node status (the agent
+        publishes elasticgpu.io/gpu-core + gpu-memory capacity)."""
+        raise NotImplementedError
+
     # events (the reference wires a recorder but never emits — controller.go:57-65;
     # we actually emit scheduling events)
     def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
@@ -231,6 +237,18 @@ class FakeKubeClient(KubeClient):
             node["metadata"]["resourceVersion"] = self._next_rv()
             return _jcopy(node)
 
+    def patch_node_allocatable(self, name: str,
+                               allocatable: Dict[str, str]) -> Node:
+        with self._mu:
+            if name not in self._nodes:
+                raise NotFoundError(f"node {name} not found")
+            node = self._nodes[name]
+            status = node.setdefault("status", {})
+            status.setdefault("allocatable", {}).update(allocatable)
+            status.setdefault("capacity", {}).update(allocatable)
+            node["metadata"]["resourceVersion"] = self._next_rv()
+            return _jcopy(node)
+
     def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
         with self._mu:
             self._events.append(_jcopy(event))
@@ -382,6 +400,14 @@ class RealKubeClient(KubeClient):
         patch = {"metadata": {"annotations": annotations}}
         return self._check(self._client.patch(
             f"/api/v1/nodes/{name}", content=json.dumps(patch),
+            headers={"Content-Type": "application/strategic-merge-patch+json"}))
+
+    def patch_node_allocatable(self, name: str,
+                               allocatable: Dict[str, str]) -> Node:
+        patch = {"status": {"allocatable": allocatable,
+                            "capacity": allocatable}}
+        return self._check(self._client.patch(
+            f"/api/v1/nodes/{name}/status", content=json.dumps(patch),
             headers={"Content-Type": "application/strategic-merge-patch+json"}))
 
     def create_event(self, namespace: str, event: Dict[str, Any]) -> None:

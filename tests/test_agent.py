@@ -120,3 +120,32 @@ def test_parse_compute_partition():
                           "system": {"driver": "x"}})
     assert inv.parse_compute_partition(payload) == {0: "SPX", 1: "CPX"}
     assert inv.parse_compute_partition("junk") == {}
+
+
+def test_parse_amd_smi_static():
+    payload = json.dumps([
+        {"gpu": 0, "asic": {"market_name": "MI355X",
+                            "target_graphics_version": "gfx950"},
+         "vram": {"size": {"value": 294912, "unit": "MB"}}},
+        {"gpu": 1, "vram": {"size": {"value": 288, "unit": "GB"}}},
+    ])
+    cards = inv.parse_amd_smi_static(payload)
+    assert len(cards) == 2
+    assert cards[0]["memory_bytes"] == 294912 * 1024**2
+    assert cards[0]["name"] == "MI355X"
+    assert cards[1]["memory_bytes"] == 288 * 1024**3
+    assert inv.parse_amd_smi_static("garbage") == []
+
+
+def test_agent_publish_includes_allocatable(monkeypatch):
+    cards = [{"index": i, "memory_bytes": 288 * GiB, "core": 100}
+             for i in range(8)]
+    monkeypatch.setattr(inv, "discover", lambda prefer="auto": cards)
+    monkeypatch.setattr(topo, "discover",
+                        lambda n, prefer="auto": topo.default_hive(n))
+    client = FakeKubeClient()
+    client.add_node({"metadata": {"name": "gpu-node"}, "status": {}})
+    NodeAgent("gpu-node", client).publish()
+    node = client.get_node("gpu-node")
+    assert node["status"]["allocatable"]["elasticgpu.io/gpu-core"] == "800"
+    assert node["status"]["capacity"]["amd.com/gpu"] == "8"
