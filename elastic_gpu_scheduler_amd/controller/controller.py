@@ -67,6 +67,17 @@ class Controller:
         for t in self._threads:
             t.join(timeout=2.0)
 
+    def wait_idle(self, timeout: float = 5.0) -> bool:
+        """Wait until the workqueue has drained (tests / shutdown)."""
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            with self._pending_mu:
+                empty = not self._pending
+            if empty and self._queue.empty():
+                return True
+            time.sleep(0.01)
+        return False
+
     # -- watch handling (reference controller.go:212-299) --
 
     def _on_event(self, event_type: str, pod: Dict[str, Any]) -> None:

@@ -100,6 +100,7 @@ def test_resync_evicts_vanished_pods():
         pod = client.create_pod(make_pod("p", core=40, memory=GiB))
         sch.assume(["n1"], pod)
         sch.bind("n1", pod)
+        assert ctrl.wait_idle()  # drain the bind's own MODIFIED events first
         # vanish without a DELETE event: remove directly from the store
         with client._mu:
             client._pods.clear()
