@@ -135,6 +135,9 @@ PYBIND11_MODULE(_core, m) {
            py::call_guard<py::gil_scoped_release>())
       .def("known_pod", &ClusterState::known_pod,
            py::call_guard<py::gil_scoped_release>())
+      .def("feasible_with_victims", &ClusterState::feasible_with_victims,
+           py::arg("node"), py::arg("uid"), py::arg("request"),
+           py::arg("victims"), py::call_guard<py::gil_scoped_release>())
       .def("node_devices",
            [](ClusterState& cs, const std::string& name) {
              auto alloc = cs.get(name);

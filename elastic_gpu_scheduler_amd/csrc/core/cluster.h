@@ -231,6 +231,14 @@ class ClusterState {
     return pod_node_.count(uid) > 0;
   }
 
+  bool feasible_with_victims(const std::string& node, const std::string& /*uid*/,
+                             const GPURequest& req,
+                             const std::vector<std::string>& victims) {
+    auto alloc = get(node);
+    if (!alloc) return false;
+    return alloc->feasible_with_victims(req, victims, *rater_);
+  }
+
   std::shared_ptr<NodeAllocator> get(const std::string& name) {
     std::shared_lock<std::shared_mutex> g(mu_);
     auto it = nodes_.find(name);

@@ -123,6 +123,40 @@ class NodeAllocator {
     return pods_.count(uid) > 0;
   }
 
+  // What-if for preemption: would `req` fit if the given victim pods were
+  // evicted? Pure read — nothing is committed. Victim UIDs not accounted on
+  // this node are ignored.
+  bool feasible_with_victims(const GPURequest& req,
+                             const std::vector<std::string>& victims,
+                             const Rater& rater) {
+    std::lock_guard<std::mutex> g(mu_);
+    std::vector<Device> copy = devices_;
+    for (const auto& uid : victims) {
+      auto it = pods_.find(uid);
+      if (it == pods_.end()) continue;
+      const GPURequest& vreq = requests_[uid];
+      const GPUOption& opt = it->second;
+      for (size_t c = 0; c < opt.allocated.size() && c < vreq.size(); ++c) {
+        const GPUUnit& u = vreq[c];
+        for (int idx : opt.allocated[c]) {
+          Device& d = copy[idx];
+          if (u.whole_cards()) {
+            d.core_avail = d.core_total;
+            d.mem_avail = d.mem_total;
+          } else {
+            d.core_avail = std::min(d.core_total, d.core_avail + u.core);
+            d.mem_avail = std::min(d.mem_total, d.mem_avail + u.memory);
+          }
+        }
+      }
+    }
+    RateContext ctx;
+    ctx.devices = &copy;
+    ctx.topo = &topo_;
+    ctx.salt = detail::fnv1a(1469598103ULL, std::hash<std::string>{}(name_));
+    return search_placement(copy, req, rater, ctx).feasible;
+  }
+
   std::vector<Device> snapshot() {
     std::lock_guard<std::mutex> g(mu_);
     return devices_;

@@ -242,6 +242,32 @@ class GPUUnitScheduler:
         with self._released_mu:
             return obj.pod_uid(pod) in self._released
 
+    def process_preemption(self, pod: Dict[str, Any],
+                           node_to_victims: Dict[str, List[str]]
+                           ) -> Dict[str, List[str]]:
+        """Extender preemption: for each candidate node, decide whether
+        evicting (a minimal subset of) the proposed victim pods makes this
+        pod feasible. Returns node -> victim UIDs actually required; nodes
+        where even all victims don't help are omitted. Capability beyond the
+        reference (neither it nor this verb exist upstream)."""
+        req = obj.pod_gpu_request(pod, self.bare_unit)
+        uid = obj.pod_uid(pod)
+        out: Dict[str, List[str]] = {}
+        for node, victims in node_to_victims.items():
+            if not self._ensure_node(node):
+                continue
+            victims = sorted(set(victims))  # deterministic
+            if not self.state.feasible_with_victims(node, uid, req, victims):
+                continue  # even evicting everything doesn't fit
+            # greedy minimisation: drop victims that aren't needed
+            needed = list(victims)
+            for v in victims:
+                trial = [x for x in needed if x != v]
+                if self.state.feasible_with_victims(node, uid, req, trial):
+                    needed = trial
+            out[node] = needed
+        return out
+
     # ---- observability ---------------------------------------------------
 
     def status(self) -> Dict[str, Any]:

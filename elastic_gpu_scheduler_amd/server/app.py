@@ -44,6 +44,7 @@ class ExtenderApp:
             ("POST", "/scheduler/filter"): self.filter,
             ("POST", "/scheduler/priorities"): self.priorities,
             ("POST", "/scheduler/bind"): self.bind,
+            ("POST", "/scheduler/preemption"): self.preemption,
             ("GET", "/scheduler/status"): self.status,
             ("GET", "/version"): self.version,
             ("GET", "/metrics"): self.metrics,
@@ -169,6 +170,32 @@ class ExtenderApp:
             if dt >= 0:
                 metrics.FILTER_TO_BIND.observe(dt)
         return 200, {}, None
+
+    def preemption(self, body: bytes):
+        """ExtenderPreemptionArgs -> ExtenderPreemptionResult. Accepts both
+        nodeNameToVictims (full pods) and nodeNameToMetaVictims (UIDs, the
+        nodeCacheCapable form); always answers with MetaVictims."""
+        args = _parse_json(body)
+        pod = args.get("pod")
+        if not pod:
+            raise _BadRequest("ExtenderPreemptionArgs.pod missing")
+        node_to_victims = {}
+        meta = args.get("nodeNameToMetaVictims") or {}
+        for node, victims in meta.items():
+            node_to_victims[node] = [p.get("uid", "")
+                                     for p in (victims or {}).get("pods", [])]
+        full = args.get("nodeNameToVictims") or {}
+        for node, victims in full.items():
+            node_to_victims.setdefault(node, []).extend(
+                obj.pod_uid(p) for p in (victims or {}).get("pods", []))
+        sch = self.registry.for_pod(pod)
+        if sch is None:
+            return 200, {"nodeNameToMetaVictims": {}}, None
+        result = sch.process_preemption(pod, node_to_victims)
+        return 200, {"nodeNameToMetaVictims": {
+            node: {"pods": [{"uid": u} for u in uids],
+                   "numPDBViolations": 0}
+            for node, uids in result.items()}}, None
 
     def status(self, body: bytes):
         return 200, None, self.registry.status_json().encode()
