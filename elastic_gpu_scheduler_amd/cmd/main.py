@@ -53,6 +53,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="Lease-based leader election: standby replicas wait "
                         "for the lease instead of double-booking GPUs (the "
                         "reference supports only replicas=1)")
+    p.add_argument("--tls-cert", default="", metavar="PEM",
+                   help="serve HTTPS (extender enableHTTPS) — uvicorn "
+                        "front end only")
+    p.add_argument("--tls-key", default="", metavar="PEM")
     p.add_argument("--leader-identity",
                    default=os.environ.get("POD_NAME", "") or os.uname().nodename)
     p.add_argument("--log-level", default="info")
@@ -143,6 +147,9 @@ def main(argv=None) -> int:
 
     log.info("listening on %s:%d (policy=%s mode=%s server=%s)", args.host,
              args.port, args.priority, args.mode, args.server)
+    if args.tls_cert and args.server == "native":
+        log.warning("TLS requires --server uvicorn; switching front end")
+        args.server = "uvicorn"
     if args.server == "native":
         from elastic_gpu_scheduler_amd.server.native import NativeFrontend
 
@@ -154,7 +161,9 @@ def main(argv=None) -> int:
         import uvicorn
 
         config = uvicorn.Config(app, host=args.host, port=args.port,
-                                log_level=args.log_level, access_log=False)
+                                log_level=args.log_level, access_log=False,
+                                ssl_certfile=args.tls_cert or None,
+                                ssl_keyfile=args.tls_key or None)
         server = uvicorn.Server(config)
 
         def watch_stop():

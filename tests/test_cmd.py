@@ -102,3 +102,42 @@ def test_agent_parser():
     args = agent_parser().parse_args(["--node", "n1", "--interval", "0"])
     assert args.node == "n1"
     assert args.interval == 0
+
+
+def test_serve_https_uvicorn(tmp_path):
+    """--tls-cert/--tls-key serve the extender over HTTPS (enableHTTPS)."""
+    cert, key = tmp_path / "tls.crt", tmp_path / "tls.key"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(key), "-out", str(cert), "-days", "1",
+         "-subj", "/CN=localhost"],
+        check=True, capture_output=True, timeout=120)
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "elastic_gpu_scheduler_amd.cmd.main",
+         "--fake-cluster", "1", "--port", str(port), "--host", "127.0.0.1",
+         "--server", "uvicorn", "--tls-cert", str(cert),
+         "--tls-key", str(key)],
+        cwd=str(REPO), stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        base = f"https://127.0.0.1:{port}"
+        deadline = time.time() + 60
+        ok = False
+        while time.time() < deadline:
+            try:
+                if httpx.get(base + "/healthz", verify=False,
+                             timeout=1.0).status_code == 200:
+                    ok = True
+                    break
+            except Exception:
+                time.sleep(0.1)
+        assert ok, "HTTPS server never came up"
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
