@@ -24,7 +24,7 @@ import time
 from typing import Any, Dict, Optional
 
 from elastic_gpu_scheduler_amd.k8s import objects as obj
-from elastic_gpu_scheduler_amd.k8s.client import KubeClient, NotFoundError
+from elastic_gpu_scheduler_amd.k8s.client import KubeClient
 from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
 
 log = logging.getLogger("egs.controller")
