@@ -14,6 +14,10 @@ REPO = Path(__file__).resolve().parent.parent
 def test_bench_two_rank_gloo(tmp_path):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
+    # force the CPU/gloo path even on a GPU box: two ranks sharing one GPU
+    # is not a supported RCCL topology (the driver gives each rank its own)
+    env["HIP_VISIBLE_DEVICES"] = ""
+    env["CUDA_VISIBLE_DEVICES"] = ""
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",

@@ -21,5 +21,11 @@ def test_core_clean_under_tsan(tmp_path):
     run = subprocess.run([str(binary)], capture_output=True, text=True,
                          timeout=300,
                          env={"TSAN_OPTIONS": "halt_on_error=1"})
+    if "unexpected memory mapping" in run.stderr:
+        # TSAN cannot run under this kernel's ASLR layout (needs
+        # vm.mmap_rnd_bits <= 30); not a product defect
+        import pytest
+
+        pytest.skip("ThreadSanitizer unsupported by this kernel's mmap layout")
     assert run.returncode == 0, (run.stdout[-1000:], run.stderr[-3000:])
     assert "stress ok" in run.stdout
