@@ -21,6 +21,7 @@ from pathlib import Path
 REPO = Path(__file__).resolve().parent
 PKG = REPO / "elastic_gpu_scheduler_amd"
 CORE_SRC = PKG / "csrc" / "core"
+HTTPD_SRC = PKG / "csrc" / "httpd"
 PROBE_SRC = PKG / "csrc" / "gpuprobe"
 GFX_ARCH = os.environ.get("EGS_GFX_ARCH", "gfx950")
 
@@ -52,7 +53,8 @@ def _run(cmd: list[str]) -> None:
 
 def build_core(force: bool = False) -> Path:
     out = PKG / ("_core" + _ext_suffix())
-    sources = sorted(CORE_SRC.glob("*.cc")) + sorted(CORE_SRC.glob("*.h"))
+    sources = (sorted(CORE_SRC.glob("*.cc")) + sorted(CORE_SRC.glob("*.h")) +
+               sorted(HTTPD_SRC.glob("*.cc")) + sorted(HTTPD_SRC.glob("*.h")))
     if force or _needs_rebuild(out, sources):
         cmd = (
             ["g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fvisibility=hidden",

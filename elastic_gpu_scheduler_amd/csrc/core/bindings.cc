@@ -121,11 +121,14 @@ PYBIND11_MODULE(_core, m) {
       .def("node_names", &ClusterState::node_names,
            py::call_guard<py::gil_scoped_release>())
       .def("assume", &ClusterState::assume, py::arg("nodes"), py::arg("uid"),
-           py::arg("request"), py::call_guard<py::gil_scoped_release>())
+           py::arg("request"), py::arg("distinct") = false,
+           py::call_guard<py::gil_scoped_release>())
       .def("score", &ClusterState::score, py::arg("nodes"), py::arg("uid"),
-           py::arg("request"), py::call_guard<py::gil_scoped_release>())
+           py::arg("request"), py::arg("distinct") = false,
+           py::call_guard<py::gil_scoped_release>())
       .def("allocate", &ClusterState::allocate, py::arg("node"), py::arg("uid"),
-           py::arg("request"), py::call_guard<py::gil_scoped_release>())
+           py::arg("request"), py::arg("distinct") = false,
+           py::call_guard<py::gil_scoped_release>())
       .def("add_pod", &ClusterState::add_pod, py::arg("node"), py::arg("uid"),
            py::arg("request"), py::arg("option"),
            py::call_guard<py::gil_scoped_release>())

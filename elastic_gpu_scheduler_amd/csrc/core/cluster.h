@@ -138,7 +138,8 @@ class ClusterState {
 
   // Filter fan-out: feasibility of `req` on every node in `names`.
   std::vector<int> assume(const std::vector<std::string>& names,
-                          const std::string& uid, const GPURequest& req) {
+                          const std::string& uid, const GPURequest& req,
+                          bool distinct = false) {
     std::vector<std::shared_ptr<NodeAllocator>> allocs(names.size());
     {
       std::shared_lock<std::shared_mutex> g(mu_);
@@ -150,7 +151,7 @@ class ClusterState {
     std::vector<int> verdicts(names.size(), static_cast<int>(AssumeVerdict::kUnknownNode));
     auto task = [&](int i) {
       if (!allocs[i]) return;
-      verdicts[i] = allocs[i]->assume(uid, req, *rater_)
+      verdicts[i] = allocs[i]->assume(uid, req, *rater_, distinct)
                         ? static_cast<int>(AssumeVerdict::kOk)
                         : static_cast<int>(AssumeVerdict::kInfeasible);
     };
@@ -165,7 +166,8 @@ class ClusterState {
   }
 
   std::vector<double> score(const std::vector<std::string>& names,
-                            const std::string& uid, const GPURequest& req) {
+                            const std::string& uid, const GPURequest& req,
+                            bool distinct = false) {
     std::vector<std::shared_ptr<NodeAllocator>> allocs(names.size());
     {
       std::shared_lock<std::shared_mutex> g(mu_);
@@ -177,7 +179,7 @@ class ClusterState {
     std::vector<double> scores(names.size(), kScoreMin);
     auto task = [&](int i) {
       if (!allocs[i]) return;
-      scores[i] = allocs[i]->score(uid, req, *rater_);
+      scores[i] = allocs[i]->score(uid, req, *rater_, distinct);
     };
     if (names.size() <= kInlineFanout) {
       for (size_t i = 0; i < names.size(); ++i) task(static_cast<int>(i));
@@ -188,10 +190,10 @@ class ClusterState {
   }
 
   GPUOption allocate(const std::string& node, const std::string& uid,
-                     const GPURequest& req) {
+                     const GPURequest& req, bool distinct = false) {
     auto alloc = get(node);
     if (!alloc) throw std::runtime_error("unknown node " + node);
-    GPUOption option = alloc->allocate(uid, req, *rater_);
+    GPUOption option = alloc->allocate(uid, req, *rater_, distinct);
     {
       std::lock_guard<std::mutex> g(pod_node_mu_);
       pod_node_[uid] = node;

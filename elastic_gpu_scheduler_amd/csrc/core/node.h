@@ -44,12 +44,13 @@ class NodeAllocator {
 
   // Feasibility + placement, cached per pod UID. Returns true if a placement
   // exists (and remembers it for the later Score/Allocate of the same pod).
-  bool assume(const std::string& uid, const GPURequest& req, const Rater& rater) {
+  bool assume(const std::string& uid, const GPURequest& req, const Rater& rater,
+              bool distinct = false) {
     std::lock_guard<std::mutex> g(mu_);
     gc_assumed_locked();
     auto it = assumed_.find(uid);
     if (it != assumed_.end()) return true;
-    auto res = run_search_locked(req, rater);
+    auto res = run_search_locked(req, rater, distinct);
     if (!res.feasible) return false;
     assumed_[uid] = {std::move(res.option), now()};
     return true;
@@ -57,11 +58,12 @@ class NodeAllocator {
 
   // Score for prioritize. A cache miss re-runs the search (never crashes on a
   // missing option). Returns kScoreMin when infeasible.
-  double score(const std::string& uid, const GPURequest& req, const Rater& rater) {
+  double score(const std::string& uid, const GPURequest& req, const Rater& rater,
+               bool distinct = false) {
     std::lock_guard<std::mutex> g(mu_);
     auto it = assumed_.find(uid);
     if (it != assumed_.end()) return it->second.option.score;
-    auto res = run_search_locked(req, rater);
+    auto res = run_search_locked(req, rater, distinct);
     if (!res.feasible) return kScoreMin;
     assumed_[uid] = {res.option, now()};
     return res.option.score;
@@ -70,7 +72,8 @@ class NodeAllocator {
   // Commit the assumed placement for a pod (bind path). If no assumed entry
   // exists (e.g. scheduler restarted between filter and bind), a fresh search
   // runs. Throws std::runtime_error when infeasible.
-  GPUOption allocate(const std::string& uid, const GPURequest& req, const Rater& rater) {
+  GPUOption allocate(const std::string& uid, const GPURequest& req, const Rater& rater,
+                     bool distinct = false) {
     std::lock_guard<std::mutex> g(mu_);
     if (pods_.count(uid)) return pods_[uid];  // idempotent re-bind
     GPUOption option;
@@ -80,12 +83,12 @@ class NodeAllocator {
       assumed_.erase(it);
       // Re-validate: the world may have changed since Assume.
       if (!fits_locked(req, option)) {
-        auto res = run_search_locked(req, rater);
+        auto res = run_search_locked(req, rater, distinct);
         if (!res.feasible) throw std::runtime_error("insufficient GPU resources on " + name_);
         option = std::move(res.option);
       }
     } else {
-      auto res = run_search_locked(req, rater);
+      auto res = run_search_locked(req, rater, distinct);
       if (!res.feasible) throw std::runtime_error("insufficient GPU resources on " + name_);
       option = std::move(res.option);
     }
@@ -186,12 +189,13 @@ class NodeAllocator {
 
   static Clock::time_point now() { return Clock::now(); }
 
-  SearchResult run_search_locked(const GPURequest& req, const Rater& rater) {
+  SearchResult run_search_locked(const GPURequest& req, const Rater& rater,
+                                 bool distinct = false) {
     RateContext ctx;
     ctx.devices = &devices_;
     ctx.topo = &topo_;
     ctx.salt = detail::fnv1a(1469598103ULL, std::hash<std::string>{}(name_));
-    return search_placement(devices_, req, rater, ctx);
+    return search_placement(devices_, req, rater, ctx, distinct);
   }
 
   bool fits_locked(const GPURequest& req, const GPUOption& option) const {

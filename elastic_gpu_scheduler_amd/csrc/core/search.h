@@ -90,6 +90,10 @@ struct DfsState {
   GPUOption best;
   bool found = false;
   int leaves = 0;
+  // distinct_containers: no card may serve two containers of this pod
+  // (elasticgpu.io/spread-containers annotation)
+  bool distinct_containers = false;
+  std::vector<bool> used_by_pod;  // card -> taken by an earlier container
 };
 
 inline void dfs(DfsState& st, size_t c);
@@ -110,12 +114,15 @@ inline void try_candidate(DfsState& st, size_t c, const std::vector<int>& cards,
       d.core_avail -= u.core;
       d.mem_avail -= u.memory;
     }
+    if (st.distinct_containers) st.used_by_pod[idx] = true;
   }
   st.current.allocated[c] = cards;
   dfs(st, c + 1);
   // Undo.
   st.current.allocated[c].clear();
   for (auto& [idx, d] : saved) st.devices[idx] = d;
+  if (st.distinct_containers)
+    for (int idx : cards) st.used_by_pod[idx] = false;
 }
 
 inline void dfs(DfsState& st, size_t c) {
@@ -141,7 +148,9 @@ inline void dfs(DfsState& st, size_t c) {
   if (u.whole_cards()) {
     std::vector<int> free_cards;
     for (int i = 0; i < static_cast<int>(st.devices.size()); ++i)
-      if (st.devices[i].whole_free()) free_cards.push_back(i);
+      if (st.devices[i].whole_free() &&
+          !(st.distinct_containers && st.used_by_pod[i]))
+        free_cards.push_back(i);
     if (static_cast<int>(free_cards.size()) < u.gpu_count) return;  // infeasible here
     std::vector<std::vector<int>> subsets;
     const Topology* topo = st.ctx->topo;
@@ -160,11 +169,12 @@ inline void dfs(DfsState& st, size_t c) {
     for (int i = 0; i < static_cast<int>(st.devices.size()); ++i) {
       const Device& d = st.devices[i];
       if (!d.can_fit(u.core, u.memory)) continue;
+      if (st.distinct_containers && st.used_by_pod[i]) continue;
       if (uniform) {
         auto sig = std::make_pair(static_cast<int64_t>(d.core_avail), d.mem_avail);
         if (std::find(seen.begin(), seen.end(), sig) != seen.end()) continue;
         seen.push_back(sig);
-      }
+      }  // (used_by_pod cards were excluded above, so dedupe stays sound)
       try_candidate(st, c, {i}, /*whole=*/false);
       if (st.leaves >= kMaxLeafEvals) return;
     }
@@ -175,12 +185,15 @@ inline void dfs(DfsState& st, size_t c) {
 
 inline SearchResult search_placement(const std::vector<Device>& devices,
                                      const GPURequest& req, const Rater& rater,
-                                     const RateContext& ctx) {
+                                     const RateContext& ctx,
+                                     bool distinct_containers = false) {
   search_detail::DfsState st;
   st.devices = devices;
   st.req = &req;
   st.rater = &rater;
   st.ctx = &ctx;
+  st.distinct_containers = distinct_containers;
+  st.used_by_pod.assign(devices.size(), false);
   st.current.allocated.resize(req.size());
   search_detail::dfs(st, 0);
   SearchResult res;

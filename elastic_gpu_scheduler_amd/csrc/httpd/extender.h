@@ -128,6 +128,7 @@ struct PodInfo {
   std::string uid;
   GPURequest request;
   bool is_gpu_pod = false;
+  bool spread_containers = false;  // elasticgpu.io/spread-containers=true
 };
 
 inline const egsjson::Value& merged_resource(const egsjson::Value& resources,
@@ -146,6 +147,11 @@ inline PodInfo parse_pod(const egsjson::Value& pod, BareUnit bare) {
 
   PodInfo info;
   info.uid = pod.get("metadata").get("uid").as_string();
+  static const std::string kSpread = "elasticgpu.io/spread-containers";
+  if (pod.get("metadata").get("annotations").get(kSpread).as_string() ==
+          "true" ||
+      pod.get("metadata").get("labels").get(kSpread).as_string() == "true")
+    info.spread_containers = true;
   const auto& containers = pod.get("spec").get("containers").as_array();
   for (const auto& c : containers) {
     const egsjson::Value& res = c.get("resources");
@@ -265,7 +271,8 @@ class ExtenderCore {
       return HandleStatus::Handled;
     }
     tracker.note(info.uid);
-    std::vector<int> verdicts = state_->assume(names, info.uid, info.request);
+    std::vector<int> verdicts =
+        state_->assume(names, info.uid, info.request, info.spread_containers);
     egsjson::Array ok;
     egsjson::Value failed = egsjson::Value::make_object();
     for (size_t i = 0; i < names.size(); ++i) {
@@ -310,7 +317,8 @@ class ExtenderCore {
         result.push_back(std::move(e));
       }
     } else {
-      std::vector<double> scores = state_->score(names, info.uid, info.request);
+      std::vector<double> scores =
+          state_->score(names, info.uid, info.request, info.spread_containers);
       for (size_t i = 0; i < names.size(); ++i) {
         egsjson::Value e = egsjson::Value::make_object();
         e.set("host", egsjson::Value(names[i]));

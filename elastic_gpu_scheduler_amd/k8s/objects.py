@@ -111,6 +111,15 @@ def pod_gpu_request(pod: Pod, bare_unit: str = "auto") -> List[Any]:
 
 # --- annotation codec (contract with the node agent) ----------------------
 
+def wants_container_spread(pod: Pod) -> bool:
+    """elasticgpu.io/spread-containers=true: place each container of this
+    pod on a distinct card."""
+    ann = pod.get("metadata", {}).get("annotations", {}) or {}
+    labels = pod.get("metadata", {}).get("labels", {}) or {}
+    return (ann.get(t.ANNOTATION_SPREAD_CONTAINERS) == "true" or
+            labels.get(t.ANNOTATION_SPREAD_CONTAINERS) == "true")
+
+
 def is_assumed(pod: Pod) -> bool:
     ann = pod.get("metadata", {}).get("annotations", {}) or {}
     return ann.get(t.EGPU_ASSUMED) == "true"

@@ -145,9 +145,10 @@ class GPUUnitScheduler:
         """Filter: returns (ok_nodes, failed_nodes: {name: reason})."""
         req = obj.pod_gpu_request(pod, self.bare_unit)
         uid = obj.pod_uid(pod)
+        distinct = obj.wants_container_spread(pod)
         for n in node_names:
             self._ensure_node(n)
-        verdicts = self.state.assume(node_names, uid, req)
+        verdicts = self.state.assume(node_names, uid, req, distinct)
         ok, failed = [], {}
         for name, v in zip(node_names, verdicts):
             if v == 0:
@@ -161,9 +162,10 @@ class GPUUnitScheduler:
     def score(self, node_names: List[str], pod: Dict[str, Any]) -> List[float]:
         req = obj.pod_gpu_request(pod, self.bare_unit)
         uid = obj.pod_uid(pod)
+        distinct = obj.wants_container_spread(pod)
         for n in node_names:
             self._ensure_node(n)
-        return self.state.score(node_names, uid, req)
+        return self.state.score(node_names, uid, req, distinct)
 
     def bind(self, node_name: str, pod: Dict[str, Any]) -> None:
         """Allocate -> annotate (conflict-retried) -> bind. Rolls back the
@@ -173,7 +175,8 @@ class GPUUnitScheduler:
         if not self._ensure_node(node_name):
             raise BindError(f"unknown or GPU-less node {node_name}")
         try:
-            option = self.state.allocate(node_name, uid, req)
+            option = self.state.allocate(node_name, uid, req,
+                                         obj.wants_container_spread(pod))
         except RuntimeError as exc:
             raise BindError(str(exc)) from exc
         try:

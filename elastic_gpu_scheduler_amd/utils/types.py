@@ -33,6 +33,11 @@ GPU_CORE_EACH_CARD = 100
 EGPU_ASSUMED = "elasticgpu.io/assumed"
 ANNOTATION_EGPU_CONTAINER_PREFIX = "elasticgpu.io/container-"
 
+# Per-pod placement constraint: "true" forces every container of the pod
+# onto DISTINCT cards (the upstream README's "spread containers of pod to
+# different GPUs" capability, which the reference never implements).
+ANNOTATION_SPREAD_CONTAINERS = "elasticgpu.io/spread-containers"
+
 # MI355X-native extensions (ours; absent on reference-scheduled pods):
 ANNOTATION_EGPU_NODE = "elasticgpu.io/scheduled-node"
 ANNOTATION_EGPU_SCORE = "elasticgpu.io/placement-score"

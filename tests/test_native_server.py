@@ -197,3 +197,28 @@ def test_native_http_edge_cases(native):
         assert c.get("/healthz").status_code == 200
         m = c.get("/metrics").content
         assert b"egs_native_requests_total" in m
+
+
+def test_native_honors_spread_containers(native):
+    """The C++ fast path must apply the spread-containers constraint too."""
+    client, registry, fe = native
+    registry.default._ensure_node("node-a")
+    pod = client.create_pod({
+        "metadata": {"name": "sp", "namespace": "default", "uid": "sp-uid",
+                     "annotations": {"elasticgpu.io/spread-containers": "true"}},
+        "spec": {"containers": [
+            {"name": f"c{i}", "resources": {"requests": {
+                "elasticgpu.io/gpu-core": "20"}}} for i in range(3)]},
+        "status": {"phase": "Pending"}})
+    with _client(fe) as c:
+        r = c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+        assert r.json()["nodenames"] == ["node-a"]
+        assert fe.stats()["filter_native"] == 1
+        r = c.post("/scheduler/bind", json={
+            "podName": "sp", "podNamespace": "default", "podUID": "sp-uid",
+            "node": "node-a"})
+        assert r.status_code == 200
+    from elastic_gpu_scheduler_amd.k8s import objects as obj
+    alloc = obj.parse_allocation(client.get_pod("default", "sp"))
+    assert len({a[0] for a in alloc}) == 3

@@ -94,3 +94,29 @@ def test_topology_locality_values():
 def test_uniform_topology_locality_is_one():
     topo = core.Topology([])
     assert topo.locality([0, 5]) == 1.0
+
+
+def test_spread_containers_constraint():
+    """elasticgpu.io/spread-containers forces distinct cards per container
+    (upstream README capability the reference never implements)."""
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(3)]
+    units = [core.GPUUnit(0, 20, GiB), core.GPUUnit(0, 20, GiB),
+             core.GPUUnit(0, 20, GiB)]
+    # unconstrained binpack packs them onto one card
+    feasible, opt, _ = core.search_placement(devs, units, "binpack", 0, [])
+    assert feasible
+    assert len({c[0] for c in opt.allocated}) == 1
+    # constrained: three distinct cards
+    c = core.ClusterState("binpack", 0, 0)
+    c.add_node("n", devs, [])
+    assert c.assume(["n"], "p", units, True) == [0]
+    o = c.allocate("n", "p", units, True)
+    cards = [x[0] for x in o.allocated]
+    assert len(set(cards)) == 3
+
+    # infeasible when cards < containers under the constraint
+    c2 = core.ClusterState("binpack", 0, 0)
+    c2.add_node("n", [core.Device(100, 100, 288 * GiB, 288 * GiB)
+                      for _ in range(2)], [])
+    assert c2.assume(["n"], "q", units, True) == [1]
+    assert c2.assume(["n"], "q2", units, False) == [0]

@@ -144,3 +144,20 @@ def test_wire_format_matches_extender_v1(cluster, extender):
     body = r.json()
     assert set(body.keys()) <= {"nodenames", "failedNodes", "nodes",
                                 "failedAndUnresolvable", "error"}
+
+
+def test_spread_containers_end_to_end(cluster, extender):
+    """A 3-container pod with elasticgpu.io/spread-containers=true lands on
+    three distinct cards, through the full HTTP pipeline."""
+    client, registry, _ = cluster
+    pod = make_pod("sp", containers=3, core=20, memory=GiB)
+    pod["metadata"]["annotations"] = {"elasticgpu.io/spread-containers": "true"}
+    created = client.create_pod(pod)
+    r = extender.filter(created, ["node-a"])
+    assert r.json()["nodenames"] == ["node-a"]
+    assert extender.bind(created, "node-a").status_code == 200
+    bound = client.get_pod("default", "sp")
+    from elastic_gpu_scheduler_amd.k8s import objects as obj
+    alloc = obj.parse_allocation(bound)
+    cards = [a[0] for a in alloc]
+    assert len(set(cards)) == 3, cards
