@@ -39,6 +39,8 @@ class Controller:
         self.resync_seconds = resync_seconds
         self._queue: "queue.Queue[Optional[str]]" = queue.Queue()
         self._pending: Dict[str, Dict[str, Any]] = {}  # key -> last seen pod
+        self._retries: Dict[str, int] = {}
+        self.max_retries = 5
         self._pending_mu = threading.Lock()
         self._stop = threading.Event()
         self._threads: list[threading.Thread] = []
@@ -109,8 +111,21 @@ class Controller:
                 continue
             try:
                 self._sync_pod(pod)
+                self._retries.pop(key, None)
             except Exception:
-                log.exception("sync of %s failed", key)
+                # rate-limited requeue (reference uses a client-go
+                # rate-limited workqueue, controller.go:64,189-210)
+                n = self._retries.get(key, 0) + 1
+                if n > self.max_retries:
+                    log.exception("sync of %s failed %d times; dropping",
+                                  key, n)
+                    self._retries.pop(key, None)
+                    continue
+                self._retries[key] = n
+                log.warning("sync of %s failed (attempt %d); requeueing",
+                            key, n, exc_info=True)
+                delay = min(0.05 * (2 ** n), 5.0)
+                threading.Timer(delay, self._enqueue, args=(key, pod)).start()
 
     def _sync_pod(self, pod: Dict[str, Any]) -> None:
         """Reference syncPod (controller.go:154-185): completed/deleted ->

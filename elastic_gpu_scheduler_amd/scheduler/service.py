@@ -281,6 +281,7 @@ class GPUUnitScheduler:
             devices = self.state.node_devices(name)
             nodes[name] = {
                 "policy": self.policy,
+                "pending_assumes": self.state.node_assumed_count(name),
                 "gpus": [{
                     "core_available": d.core_avail,
                     "core_total": d.core_total,

@@ -63,6 +63,9 @@ class ExtenderApp:
             return (404, "application/json",
                     json.dumps({"error": f"no route {method} {path}"}).encode())
         verb = path.rsplit("/", 1)[-1]
+        debug = log.isEnabledFor(logging.DEBUG)
+        if debug:
+            log.debug("%s %s body=%s", method, path, body[:2048])
         with metrics.VERB_LATENCY.labels(verb).time():
             try:
                 status, payload, raw = handler(body)
@@ -74,6 +77,9 @@ class ExtenderApp:
                 log.exception("%s %s failed", method, path)
                 status, payload, raw = 500, {"error": f"{type(exc).__name__}: {exc}"}, None
                 metrics.REQUESTS.labels(verb, "exception").inc()
+        if debug:
+            preview = raw if raw is not None else json.dumps(payload).encode()
+            log.debug("%s %s -> %d %s", method, path, status, preview[:2048])
         ctype = "text/plain; charset=utf-8" if raw is not None and \
             path in ("/debug/stacks",) else (
                 "text/plain; version=0.0.4; charset=utf-8"

@@ -125,3 +125,31 @@ def test_non_gpu_pods_ignored():
         assert not registry.default.state.has_node("n1")
     finally:
         ctrl.stop()
+
+
+def test_sync_failure_retries_with_backoff():
+    """A transient sync failure requeues the pod instead of dropping it."""
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        calls = {"n": 0}
+        orig = sch.add_pod
+
+        def flaky_add(pod):
+            calls["n"] += 1
+            if calls["n"] < 3:
+                raise RuntimeError("transient")
+            orig(pod)
+
+        sch.add_pod = flaky_add
+        pod = make_pod("ext", core=30, memory=GiB)
+        pod["metadata"]["labels"] = {"elasticgpu.io/assumed": "true"}
+        pod["metadata"]["annotations"] = {"elasticgpu.io/assumed": "true",
+                                          "elasticgpu.io/container-c0": "0"}
+        pod["spec"]["nodeName"] = "n1"
+        client.create_pod(pod)
+        assert wait_until(lambda: any(
+            d.core_avail == 70 for d in sch.state.node_devices("n1")))
+        assert calls["n"] >= 3
+    finally:
+        ctrl.stop()
