@@ -155,13 +155,7 @@ class ClusterState {
                         ? static_cast<int>(AssumeVerdict::kOk)
                         : static_cast<int>(AssumeVerdict::kInfeasible);
     };
-    // Small fan-outs run inline: per-node search is ~5-10 us, so pool
-    // dispatch overhead dominates below a few dozen nodes.
-    if (names.size() <= kInlineFanout) {
-      for (size_t i = 0; i < names.size(); ++i) task(static_cast<int>(i));
-    } else {
-      pool_.parallel_for(static_cast<int>(names.size()), task);
-    }
+    run_fanout(names.size(), task);
     return verdicts;
   }
 
@@ -181,11 +175,7 @@ class ClusterState {
       if (!allocs[i]) return;
       scores[i] = allocs[i]->score(uid, req, *rater_, distinct);
     };
-    if (names.size() <= kInlineFanout) {
-      for (size_t i = 0; i < names.size(); ++i) task(static_cast<int>(i));
-    } else {
-      pool_.parallel_for(static_cast<int>(names.size()), task);
-    }
+    run_fanout(names.size(), task);
     return scores;
   }
 
@@ -241,6 +231,25 @@ class ClusterState {
     return alloc->feasible_with_victims(req, victims, *rater_);
   }
 
+  // Adaptive fan-out: small node counts run inline (pool dispatch costs
+  // more than the searches); large fan-outs use the pool ONLY when few
+  // requests are in flight — concurrent requests are already parallel
+  // across server threads, and stacking their fan-outs onto one shared
+  // pool just thrashes its queue lock.
+  void run_fanout(size_t n, const std::function<void(int)>& task) {
+    struct Guard {
+      std::atomic<int>& c;
+      explicit Guard(std::atomic<int>& c_) : c(c_) { c.fetch_add(1); }
+      ~Guard() { c.fetch_sub(1); }
+    } guard(inflight_);
+    bool use_pool = n > kInlineFanout && inflight_.load() <= 2;
+    if (use_pool) {
+      pool_.parallel_for(static_cast<int>(n), task);
+    } else {
+      for (size_t i = 0; i < n; ++i) task(static_cast<int>(i));
+    }
+  }
+
   std::shared_ptr<NodeAllocator> get(const std::string& name) {
     std::shared_lock<std::shared_mutex> g(mu_);
     auto it = nodes_.find(name);
@@ -252,6 +261,7 @@ class ClusterState {
  private:
   std::unique_ptr<Rater> rater_;
   ThreadPool pool_;
+  std::atomic<int> inflight_{0};
   std::shared_mutex mu_;
   std::unordered_map<std::string, std::shared_ptr<NodeAllocator>> nodes_;
   std::mutex pod_node_mu_;
