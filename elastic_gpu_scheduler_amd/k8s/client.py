@@ -433,6 +433,10 @@ class RealKubeClient(KubeClient):
         stop = threading.Event()
 
         def run() -> None:
+            # Reconnects restart the watch at "now" (no resourceVersion
+            # bookmark), so events during a gap are missed — the
+            # controller's periodic resync (Controller.resync_once) is the
+            # drift repair, same division of labor as informer relists.
             while not stop.is_set():
                 try:
                     with self._client.stream(
