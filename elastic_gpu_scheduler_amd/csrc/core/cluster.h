@@ -11,6 +11,7 @@
 
 #include <atomic>
 #include <condition_variable>
+#include <deque>
 #include <functional>
 #include <memory>
 #include <shared_mutex>
@@ -85,13 +86,13 @@ class ThreadPool {
         cv_.wait(g, [this] { return stop_ || !queue_.empty(); });
         if (stop_ && queue_.empty()) return;
         task = std::move(queue_.front());
-        queue_.erase(queue_.begin());
+        queue_.pop_front();
       }
       task();
     }
   }
   std::vector<std::thread> workers_;
-  std::vector<std::function<void()>> queue_;
+  std::deque<std::function<void()>> queue_;
   std::mutex mu_;
   std::condition_variable cv_;
   bool stop_ = false;
