@@ -44,6 +44,9 @@ def build_parser() -> argparse.ArgumentParser:
                    choices=("auto", "bytes", "GiB", "MiB"),
                    help="how to read suffix-less gpu-memory quantities")
     p.add_argument("--seed", type=int, default=0, help="random-policy seed")
+    p.add_argument("--topology-weight", type=float, default=0.3,
+                   help="xGMI-locality share of a multi-card placement's "
+                        "score, 0..1 (0 disables topology steering)")
     p.add_argument("--fake-cluster", type=int, default=0, metavar="NODES",
                    help="serve against an in-process fake apiserver with N "
                         "8x-MI355X nodes (local dev / benchmarking)")
@@ -125,7 +128,8 @@ def main(argv=None) -> int:
 
     registry = SchedulerRegistry(client, mode=args.mode, policy=args.priority,
                                  seed=args.seed, threads=args.filter_threads,
-                                 bare_unit=args.bare_memory_unit)
+                                 bare_unit=args.bare_memory_unit,
+                                 topology_weight=args.topology_weight)
     controller = Controller(client, registry, workers=args.threadness)
     controller.start()
 

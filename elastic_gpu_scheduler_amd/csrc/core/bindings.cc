@@ -108,8 +108,9 @@ PYBIND11_MODULE(_core, m) {
   bind_native_server(m);
 
   py::class_<ClusterState, std::shared_ptr<ClusterState>>(m, "ClusterState")
-      .def(py::init<const std::string&, uint64_t, int>(), py::arg("policy") = "binpack",
-           py::arg("seed") = 0, py::arg("threads") = 0)
+      .def(py::init<const std::string&, uint64_t, int, double>(),
+           py::arg("policy") = "binpack", py::arg("seed") = 0,
+           py::arg("threads") = 0, py::arg("topology_weight") = 0.3)
       .def_property_readonly("policy", &ClusterState::policy)
       .def_property_readonly("pool_size", &ClusterState::pool_size)
       .def("add_node", &ClusterState::add_node, py::arg("name"), py::arg("devices"),

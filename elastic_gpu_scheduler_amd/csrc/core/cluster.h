@@ -115,15 +115,18 @@ enum class AssumeVerdict : int {
 
 class ClusterState {
  public:
-  ClusterState(const std::string& policy, uint64_t seed, int threads)
-      : rater_(make_rater(policy, seed)), pool_(threads) {}
+  ClusterState(const std::string& policy, uint64_t seed, int threads,
+               double topology_weight = kDefaultTopologyWeight)
+      : rater_(make_rater(policy, seed)), pool_(threads),
+        topology_weight_(topology_weight) {}
 
   std::string policy() const { return rater_->name(); }
 
   void add_node(const std::string& name, std::vector<Device> devices,
                 std::vector<std::vector<int>> topo_hops) {
     auto alloc = std::make_shared<NodeAllocator>(name, std::move(devices),
-                                                Topology(std::move(topo_hops)));
+                                                Topology(std::move(topo_hops)),
+                                                topology_weight_);
     std::unique_lock<std::shared_mutex> g(mu_);
     nodes_[name] = std::move(alloc);  // replaces any stale entry
   }
@@ -274,6 +277,7 @@ class ClusterState {
  private:
   std::unique_ptr<Rater> rater_;
   ThreadPool pool_;
+  double topology_weight_ = kDefaultTopologyWeight;
   std::atomic<int> inflight_{0};
   std::shared_mutex mu_;
   std::unordered_map<std::string, std::shared_ptr<NodeAllocator>> nodes_;

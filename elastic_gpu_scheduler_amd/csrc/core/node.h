@@ -36,8 +36,10 @@ constexpr auto kAssumeTTL = std::chrono::seconds(300);
 
 class NodeAllocator {
  public:
-  NodeAllocator(std::string name, std::vector<Device> devices, Topology topo)
-      : name_(std::move(name)), devices_(std::move(devices)), topo_(std::move(topo)) {}
+  NodeAllocator(std::string name, std::vector<Device> devices, Topology topo,
+                double topology_weight = kDefaultTopologyWeight)
+      : name_(std::move(name)), devices_(std::move(devices)),
+        topo_(std::move(topo)), topology_weight_(topology_weight) {}
 
   const std::string& name() const { return name_; }
   int num_devices() const { return static_cast<int>(devices_.size()); }
@@ -156,6 +158,7 @@ class NodeAllocator {
     RateContext ctx;
     ctx.devices = &copy;
     ctx.topo = &topo_;
+    ctx.topology_weight = topology_weight_;
     ctx.salt = detail::fnv1a(1469598103ULL, std::hash<std::string>{}(name_));
     return search_placement(copy, req, rater, ctx).feasible;
   }
@@ -194,6 +197,7 @@ class NodeAllocator {
     RateContext ctx;
     ctx.devices = &devices_;
     ctx.topo = &topo_;
+    ctx.topology_weight = topology_weight_;
     ctx.salt = detail::fnv1a(1469598103ULL, std::hash<std::string>{}(name_));
     return search_placement(devices_, req, rater, ctx, distinct);
   }
@@ -285,6 +289,7 @@ class NodeAllocator {
   std::string name_;
   std::vector<Device> devices_;
   Topology topo_;
+  double topology_weight_ = kDefaultTopologyWeight;
   std::mutex mu_;
   std::unordered_map<std::string, Assumed> assumed_;      // uid -> pending placement
   std::unordered_map<std::string, GPUOption> pods_;       // uid -> committed placement

@@ -22,13 +22,15 @@ namespace egs {
 constexpr double kScoreMin = 0.0;
 constexpr double kScoreMax = 10.0;
 
-// Weight of the xGMI-locality term for multi-card containers.
-constexpr double kTopologyWeight = 0.3;
+// Default weight of the xGMI-locality term for multi-card containers
+// (operators tune via --topology-weight).
+constexpr double kDefaultTopologyWeight = 0.3;
 
 struct RateContext {
   const std::vector<Device>* devices = nullptr;  // state BEFORE the placement
   const Topology* topo = nullptr;
   uint64_t salt = 0;  // stable per (node, pod): differentiates Random scores
+  double topology_weight = kDefaultTopologyWeight;
 };
 
 namespace detail {
@@ -94,8 +96,9 @@ inline double topology_locality(const RateContext& ctx, const GPUOption& option)
 // Blend a base policy score with xGMI locality for multi-card placements.
 inline double blend(const RateContext& ctx, const GPUOption& option, double base) {
   double loc = topology_locality(ctx, option);
-  if (loc < 0.0) return std::clamp(base, kScoreMin, kScoreMax);
-  double s = (1.0 - kTopologyWeight) * base + kTopologyWeight * kScoreMax * loc;
+  double w = std::clamp(ctx.topology_weight, 0.0, 1.0);
+  if (loc < 0.0 || w == 0.0) return std::clamp(base, kScoreMin, kScoreMax);
+  double s = (1.0 - w) * base + w * kScoreMax * loc;
   return std::clamp(s, kScoreMin, kScoreMax);
 }
 

@@ -43,12 +43,12 @@ class GPUUnitScheduler:
 
     def __init__(self, client: KubeClient, policy: str = t.PRIORITY_BINPACK,
                  seed: int = 0, threads: int = 0, bare_unit: str = "auto",
-                 name: str = "gpushare") -> None:
+                 name: str = "gpushare", topology_weight: float = 0.3) -> None:
         self.client = client
         self.name = name
         self.policy = policy
         self.bare_unit = bare_unit
-        self.state = core.ClusterState(policy, seed, threads)
+        self.state = core.ClusterState(policy, seed, threads, topology_weight)
         # released-pod tombstones (reference releasedPodMap, scheduler.go:47):
         # a DELETE seen before the final MODIFIED must not re-add the pod.
         self._released: Dict[str, float] = {}
@@ -340,10 +340,12 @@ class SchedulerRegistry:
 
     def __init__(self, client: KubeClient, mode: str = t.MODE_GPUSHARE,
                  policy: str = t.PRIORITY_BINPACK, seed: int = 0,
-                 threads: int = 0, bare_unit: str = "auto") -> None:
+                 threads: int = 0, bare_unit: str = "auto",
+                 topology_weight: float = 0.3) -> None:
         self.schedulers: Dict[str, GPUUnitScheduler] = {}
         unit = GPUUnitScheduler(client, policy=policy, seed=seed, threads=threads,
-                                bare_unit=bare_unit, name=mode)
+                                bare_unit=bare_unit, name=mode,
+                                topology_weight=topology_weight)
         if mode == t.MODE_QGPU:
             names = (t.RESOURCE_QGPU_CORE, t.RESOURCE_QGPU_MEMORY)
         elif mode == t.MODE_PGPU:

@@ -120,3 +120,24 @@ def test_spread_containers_constraint():
                       for _ in range(2)], [])
     assert c2.assume(["n"], "q", units, True) == [1]
     assert c2.assume(["n"], "q2", units, False) == [0]
+
+
+def test_topology_weight_configurable():
+    hops = [[0 if i == j else (1 if (i < 4) == (j < 4) else 3)
+             for j in range(8)] for i in range(8)]
+    # weight 0: topology ignored entirely -> scores equal the pure policy
+    c0 = core.ClusterState("binpack", 0, 0, 0.0)
+    c0.add_node("n", devices(8), hops)
+    c9 = core.ClusterState("binpack", 0, 0, 0.9)
+    c9.add_node("n", devices(8), hops)
+    # occupy card 5 slightly so a cross-hive pair can win on pure binpack
+    for c in (c0, c9):
+        c.allocate("n", "warm", frac(10, GiB))
+    req = [core.GPUUnit(0, 0, 0)]  # placeholder; use 2 whole cards:
+    req = [core.GPUUnit(2, 0, 0)]
+    o9 = c9.allocate("n", "p", req)
+    cards9 = o9.allocated[0]
+    assert len({i < 4 for i in cards9}) == 1  # heavy weight: stays in-hive
+    s0 = c0.score(["n"], "q", req)[0]
+    s9 = c9.score(["n"], "q", req)[0]
+    assert 0 <= s0 <= 10 and 0 <= s9 <= 10
