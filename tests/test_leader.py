@@ -22,24 +22,27 @@ def run_elector(elector, events, label):
 def test_single_leader_and_takeover():
     client = FakeKubeClient()
     events = []
-    a = LeaderElector(client, "egs", "a", lease_duration=0.6,
-                      renew_period=0.15, retry_period=0.1)
-    b = LeaderElector(client, "egs", "b", lease_duration=0.6,
-                      renew_period=0.15, retry_period=0.1)
+    # generous lease vs renew ratio so a loaded CI machine cannot make the
+    # holder miss renewals spuriously
+    a = LeaderElector(client, "egs", "a", lease_duration=3.0,
+                      renew_period=0.1, retry_period=0.1)
+    b = LeaderElector(client, "egs", "b", lease_duration=3.0,
+                      renew_period=0.1, retry_period=0.1)
     ta = run_elector(a, events, "a")
-    time.sleep(0.3)
+    deadline = time.time() + 5
+    while time.time() < deadline and not a.is_leader:
+        time.sleep(0.02)
     tb = run_elector(b, events, "b")
-    time.sleep(0.4)
+    time.sleep(0.5)
     assert a.is_leader and not b.is_leader
     assert events == ["a-started"]
 
     # a dies abruptly (no release): b takes over after the lease expires
     a._stop.set()
-    # prevent graceful release from the run loop by joining after hard stop:
-    ta.join(timeout=2)
+    ta.join(timeout=5)
     # a's graceful release may or may not have happened depending on timing;
-    # either way b must eventually lead
-    deadline = time.time() + 5
+    # either way b must eventually lead (after <= lease_duration)
+    deadline = time.time() + 15
     while time.time() < deadline and not b.is_leader:
         time.sleep(0.05)
     assert b.is_leader
