@@ -14,6 +14,7 @@
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/socket.h>
+#include <sys/time.h>
 #include <unistd.h>
 
 #include <atomic>
@@ -113,6 +114,12 @@ class HttpServer {
       }
       int one = 1;
       setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      // Bound how long a half-open client can pin a connection thread:
+      // idle keep-alive connections are closed after this and clients
+      // (kube-scheduler) reconnect transparently.
+      timeval tv{300, 0};
+      setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+      setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
       {
         std::lock_guard<std::mutex> g(conn_mu_);
         if (static_cast<int>(conn_fds_.size()) >= max_connections_) {

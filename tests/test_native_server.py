@@ -222,3 +222,48 @@ def test_native_honors_spread_containers(native):
     from elastic_gpu_scheduler_amd.k8s import objects as obj
     alloc = obj.parse_allocation(client.get_pod("default", "sp"))
     assert len({a[0] for a in alloc}) == 3
+
+
+def test_native_connection_churn_and_partial_requests(native):
+    """Abrupt disconnects, partial requests, and many short-lived
+    connections must not wedge or leak the server."""
+    import socket as _socket
+    import threading
+
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("p", core=10, memory=GiB))
+    registry.default.assume(["node-a"], pod)
+    body = json.dumps({"pod": pod, "nodenames": ["node-a"]}).encode()
+    payload = (b"POST /scheduler/filter HTTP/1.1\r\nhost: t\r\n"
+               b"content-length: " + str(len(body)).encode() + b"\r\n\r\n" + body)
+
+    def churn(kind):
+        for _ in range(30):
+            s = _socket.create_connection(("127.0.0.1", fe.port))
+            try:
+                if kind == "abrupt":
+                    s.sendall(payload[: len(payload) // 2])  # die mid-request
+                elif kind == "headers-only":
+                    s.sendall(b"POST /scheduler/filter HTTP/1.1\r\n"
+                              b"content-length: 100\r\n\r\n")  # body never comes
+                else:
+                    s.sendall(payload)
+                    data = b""
+                    while b"\r\n\r\n" not in data:
+                        data += s.recv(65536)
+            finally:
+                s.close()
+
+    threads = [threading.Thread(target=churn, args=(k,))
+               for k in ("abrupt", "headers-only", "full", "full")]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not any(t.is_alive() for t in threads)
+    # server still healthy afterwards
+    with _client(fe) as c:
+        assert c.get("/healthz").status_code == 200
+        r = c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+        assert r.status_code == 200
