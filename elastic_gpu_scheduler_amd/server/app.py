@@ -50,6 +50,7 @@ class ExtenderApp:
             ("GET", "/metrics"): self.metrics,
             ("GET", "/healthz"): self.healthz,
             ("GET", "/debug/stacks"): self.debug_stacks,
+            ("GET", "/debug/profile"): self.debug_profile,
         }
 
     # ---- shared sync dispatcher -----------------------------------------
@@ -81,7 +82,7 @@ class ExtenderApp:
             preview = raw if raw is not None else json.dumps(payload).encode()
             log.debug("%s %s -> %d %s", method, path, status, preview[:2048])
         ctype = "text/plain; charset=utf-8" if raw is not None and \
-            path in ("/debug/stacks",) else (
+            path in ("/debug/stacks", "/debug/profile") else (
                 "text/plain; version=0.0.4; charset=utf-8"
                 if path == "/metrics" else "application/json")
         return (status, ctype,
@@ -226,6 +227,32 @@ class ExtenderApp:
 
     def healthz(self, body: bytes):
         return 200, {"ok": True}, None
+
+    def debug_profile(self, body: bytes):
+        """Sampling CPU profile of the Python threads (analogue of
+        /debug/pprof/profile, pkg/routes/pprof.go): samples all thread
+        stacks for ~2 s at 100 Hz and returns collapsed stacks (one
+        `frame;frame;frame count` per line — feed to a flamegraph tool).
+        The C++ fast-path threads are GIL-free and invisible here."""
+        import collections
+        import time as _time
+
+        samples: "collections.Counter[str]" = collections.Counter()
+        deadline = _time.time() + 2.0
+        while _time.time() < deadline:
+            for tid, frame in sys._current_frames().items():
+                frames = []
+                f = frame
+                while f is not None and len(frames) < 64:
+                    code = f.f_code
+                    frames.append(f"{code.co_filename.rsplit('/', 1)[-1]}:"
+                                  f"{code.co_name}")
+                    f = f.f_back
+                samples[";".join(reversed(frames))] += 1
+            _time.sleep(0.01)
+        out = "\n".join(f"{stack} {count}"
+                         for stack, count in samples.most_common())
+        return 200, None, (out + "\n").encode()
 
     def debug_stacks(self, body: bytes):
         frames = sys._current_frames()

@@ -161,3 +161,14 @@ def test_spread_containers_end_to_end(cluster, extender):
     alloc = obj.parse_allocation(bound)
     cards = [a[0] for a in alloc]
     assert len(set(cards)) == 3, cards
+
+
+def test_debug_profile_collapsed_stacks(cluster, extender):
+    r = extender.request("GET", "/debug/profile")
+    assert r.status_code == 200
+    text = r.text
+    # every line is "frame;frame;... count"
+    line = text.strip().splitlines()[0]
+    stack, _, count = line.rpartition(" ")
+    assert int(count) >= 1
+    assert ":" in stack
