@@ -29,3 +29,22 @@ def test_core_clean_under_tsan(tmp_path):
         pytest.skip("ThreadSanitizer unsupported by this kernel's mmap layout")
     assert run.returncode == 0, (run.stdout[-1000:], run.stderr[-3000:])
     assert "stress ok" in run.stdout
+
+
+def test_core_clean_under_asan(tmp_path):
+    """Same stress harness under AddressSanitizer + UBSan (heap errors,
+    overflow, UB in the allocator core)."""
+    binary = tmp_path / "stress_asan"
+    build = subprocess.run(
+        ["g++", "-O1", "-g", "-std=c++17",
+         "-fsanitize=address,undefined", "-fno-sanitize-recover=all",
+         "-pthread",
+         "-I", str(REPO / "elastic_gpu_scheduler_amd/csrc/core"),
+         str(REPO / "elastic_gpu_scheduler_amd/csrc/stress/stress_main.cc"),
+         "-o", str(binary)],
+        capture_output=True, text=True, timeout=300)
+    assert build.returncode == 0, build.stderr[-3000:]
+    run = subprocess.run([str(binary)], capture_output=True, text=True,
+                         timeout=300)
+    assert run.returncode == 0, (run.stdout[-1000:], run.stderr[-3000:])
+    assert "stress ok" in run.stdout
