@@ -42,7 +42,7 @@ GiB = 1024**3
 
 # ---------------------------------------------------------------- workload
 
-def mixed_pod_spec(i: int, step: int, rank: int):
+def mixed_pod_spec(i: int, step: int, rank: int, pad_bytes: int = 0):
     """BASELINE config #4 mix: whole-card, half-card, quarter and small
     fractional pods, deterministic by index."""
     k = i % 8
@@ -58,7 +58,7 @@ def mixed_pod_spec(i: int, step: int, rank: int):
     req = {"elasticgpu.io/gpu-core": str(core)}
     if mem:
         req["elasticgpu.io/gpu-memory"] = str(mem)
-    return {
+    pod = {
         "metadata": {"name": name, "namespace": "default",
                      "uid": str(uuid.uuid4())},
         "spec": {"containers": [{"name": "main",
@@ -66,6 +66,11 @@ def mixed_pod_spec(i: int, step: int, rank: int):
                                                "limits": dict(req)}}]},
         "status": {"phase": "Pending"},
     }
+    if pad_bytes > 0:
+        # realistic pod objects carry kB of env/volumes/labels; model that
+        # wire weight with an opaque annotation
+        pod["metadata"]["annotations"] = {"bench.pad": "x" * pad_bytes}
+    return pod
 
 
 def build_cluster(n_nodes: int, cards: int, use_gpu_inventory: bool):
@@ -231,7 +236,9 @@ class BenchPipeline:
     # -- one step --
 
     def step(self, step_idx: int, record_latency: bool):
-        pods = [self.client.create_pod(mixed_pod_spec(i, step_idx, self.rank))
+        pods = [self.client.create_pod(
+                    mixed_pod_spec(i, step_idx, self.rank,
+                                   self.args.pod_pad_bytes))
                 for i in range(self.args.batch)]
         if self.args.no_http:
             self._schedule_direct(pods, record_latency)
@@ -398,6 +405,9 @@ def main():
                    help="bypass TCP; drive handlers in-process")
     p.add_argument("--no-verify", action="store_true",
                    help="skip on-GPU placement stamping")
+    p.add_argument("--pod-pad-bytes", type=int, default=0,
+                   help="pad each pod object with N annotation bytes to "
+                        "model realistic (multi-kB) pod specs on the wire")
     p.add_argument("--verify-sample", type=int, default=2,
                    help="placements stamped on-device per step")
     args = p.parse_args()
@@ -484,6 +494,7 @@ def main():
                     "native_stats": (pipe._native.stats()
                                      if pipe._native else None),
                     "concurrency": args.concurrency,
+                    "pod_pad_bytes": args.pod_pad_bytes,
                     "p50_filter_bind_ms": round(p50, 3) if p50 else None,
                     "p99_filter_bind_ms": round(p99, 3) if p99 else None,
                     "verify_on_device": pipe.probe is not None,

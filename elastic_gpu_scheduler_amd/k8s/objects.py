@@ -143,11 +143,15 @@ def allocation_annotations(pod: Pod, allocated: List[List[int]],
 
 
 def apply_allocation(pod: Pod, allocated: List[List[int]], node: str = "",
-                     score: float = 0.0) -> Pod:
-    """Return a copy of the pod with placement annotations + assumed label."""
+                     score: float = 0.0, copy: bool = True) -> Pod:
+    """Return the pod with placement annotations + assumed label applied.
+
+    copy=False mutates `pod` in place — valid when the caller owns the dict
+    (e.g. a fresh get_pod result); the bind hot path uses it to skip one
+    deep copy per bind."""
     from elastic_gpu_scheduler_amd.k8s.client import _jcopy
 
-    p = _jcopy(pod)
+    p = _jcopy(pod) if copy else pod
     meta = p.setdefault("metadata", {})
     ann = meta.setdefault("annotations", {})
     ann.update(allocation_annotations(pod, allocated, node, score))

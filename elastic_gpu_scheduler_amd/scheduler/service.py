@@ -191,8 +191,10 @@ class GPUUnitScheduler:
 
     def _write_bind(self, node_name: str, pod: Dict[str, Any], option) -> None:
         ns, name = obj.pod_namespace(pod), obj.pod_name(pod)
+        # the handler fetched `pod` fresh for this bind; annotate in place
         annotated = obj.apply_allocation(pod, [list(a) for a in option.allocated],
-                                         node=node_name, score=option.score)
+                                         node=node_name, score=option.score,
+                                         copy=False)
         for attempt in range(3):
             try:
                 self.client.update_pod(annotated)
@@ -209,7 +211,7 @@ class GPUUnitScheduler:
                     raise BindError(f"pod {ns}/{name} was recreated during bind")
                 annotated = obj.apply_allocation(
                     fresh, [list(a) for a in option.allocated],
-                    node=node_name, score=option.score)
+                    node=node_name, score=option.score, copy=False)
         self.client.bind_pod(ns, name, node_name)
 
     def add_pod(self, pod: Dict[str, Any]) -> None:
