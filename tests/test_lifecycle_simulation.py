@@ -114,3 +114,32 @@ def test_lifecycle_with_restart():
         assert live, "simulation scheduled nothing"
     finally:
         ctrl.stop()
+
+
+def test_large_cluster_simulation():
+    """1,000-node cluster slice: filter fan-outs at cluster scale through
+    the service layer, placements land, accounting matches ground truth."""
+    import random
+
+    rng = random.Random(3)
+    client = FakeKubeClient()
+    nodes = [f"big{i}" for i in range(1000)]
+    for n in nodes:
+        client.add_node(make_node(n))
+    registry = SchedulerRegistry(client)
+    sch = registry.default
+    bound = []
+    for i in range(60):
+        pod = client.create_pod(make_pod(f"p{i}", core=rng.choice([25, 50]),
+                                         memory=48 * GiB))
+        ok, failed = sch.assume(nodes, pod)
+        assert len(ok) == 1000, len(ok)
+        target = rng.choice(ok)
+        sch.bind(target, pod)
+        bound.append((f"p{i}", target))
+    assert observed_usage(sch, nodes) == expected_usage(client, nodes)
+    # release half, re-check
+    for name, _ in bound[::2]:
+        sch.forget_pod(client.get_pod("default", name))
+        client.delete_pod("default", name)
+    assert observed_usage(sch, nodes) == expected_usage(client, nodes)

@@ -162,3 +162,49 @@ def test_scheduler_service_over_rest_client(api):
         in {str(i) for i in range(8)}
     sch.flush_events()
     assert mock.events, "bind must emit a scheduling event"
+
+
+def test_watch_pods_streams_and_reconnects():
+    """RealKubeClient.watch_pods: parses the apiserver's JSON-lines watch
+    stream, delivers events, and survives a dropped stream (reconnect)."""
+    import threading
+    import time
+
+    events_delivered = []
+    connections = {"n": 0}
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        if request.url.params.get("watch") == "true":
+            connections["n"] += 1
+            if connections["n"] == 1:
+                body = (json.dumps({"type": "ADDED",
+                                    "object": {"metadata": {"name": "w1"}}})
+                        + "\n" +
+                        json.dumps({"type": "MODIFIED",
+                                    "object": {"metadata": {"name": "w1"}}})
+                        + "\n")
+            else:
+                body = json.dumps({"type": "DELETED",
+                                   "object": {"metadata": {"name": "w1"}}}) + "\n"
+            return httpx.Response(200, content=body.encode())
+        return httpx.Response(404)
+
+    client = RealKubeClient("https://apiserver", token="t",
+                            transport=httpx.MockTransport(handler))
+    # shrink the reconnect backoff for the test
+    got = threading.Event()
+
+    def on_event(etype, obj):
+        events_delivered.append((etype, obj.get("metadata", {}).get("name")))
+        if etype == "DELETED":
+            got.set()
+
+    unsubscribe = client.watch_pods(on_event)
+    try:
+        assert got.wait(timeout=10), events_delivered
+    finally:
+        unsubscribe()
+    assert ("ADDED", "w1") in events_delivered
+    assert ("MODIFIED", "w1") in events_delivered
+    assert ("DELETED", "w1") in events_delivered  # arrived via reconnect
+    assert connections["n"] >= 2
