@@ -405,6 +405,9 @@ def main():
                    help="bypass TCP; drive handlers in-process")
     p.add_argument("--no-verify", action="store_true",
                    help="skip on-GPU placement stamping")
+    p.add_argument("--self-profile", default="", metavar="PATH",
+                   help="capture /debug/profile (collapsed stacks) during "
+                        "the timed steps and write it to PATH")
     p.add_argument("--pod-pad-bytes", type=int, default=0,
                    help="pad each pod object with N annotation bytes to "
                         "model realistic (multi-kB) pod specs on the wire")
@@ -443,9 +446,28 @@ def main():
         for w in range(args.warmup):
             pipe.step(-1 - w, record_latency=False)
 
+        profile_thread = None
+        if args.self_profile and rank == 0 and not args.no_http:
+            def grab_profile():
+                import urllib.request
+
+                try:
+                    with urllib.request.urlopen(
+                            pipe.base_url + "/debug/profile",
+                            timeout=30) as r:
+                        data = r.read()
+                    with open(args.self_profile, "wb") as f:
+                        f.write(data)
+                except Exception as exc:  # profiling must never fail the bench
+                    print(f"self-profile failed: {exc}", file=__import__("sys").stderr)
+
+            profile_thread = threading.Thread(target=grab_profile, daemon=True)
+
         barrier()
         t0 = time.perf_counter()
         for s in range(args.steps):
+            if s == 1 and profile_thread is not None:
+                profile_thread.start()  # sample while under steady load
             pipe.step(s, record_latency=True)
         barrier()
         elapsed = time.perf_counter() - t0
@@ -503,6 +525,9 @@ def main():
             }
             print(json.dumps(result), flush=True)
     finally:
+        if "profile_thread" in dir() and profile_thread is not None \
+                and profile_thread.is_alive():
+            profile_thread.join(timeout=30)
         pipe.close()
         if distributed:
             import torch.distributed as dist
