@@ -305,7 +305,8 @@ class BenchPipeline:
                 "node": best})
             if status == 200:
                 break
-            self.bind_retries += 1
+            with self._lat_mu:
+                self.bind_retries += 1
         else:
             raise RuntimeError(f"bind kept failing: {out}")
         if record_latency:
@@ -525,9 +526,9 @@ def main():
             }
             print(json.dumps(result), flush=True)
     finally:
-        if "profile_thread" in dir() and profile_thread is not None \
-                and profile_thread.is_alive():
-            profile_thread.join(timeout=30)
+        pt = locals().get("profile_thread")
+        if pt is not None and pt.is_alive():
+            pt.join(timeout=30)
         pipe.close()
         if distributed:
             import torch.distributed as dist

@@ -20,6 +20,7 @@
 #include <atomic>
 #include <cstring>
 #include <functional>
+#include <memory>
 #include <mutex>
 #include <stdexcept>
 #include <string>
@@ -94,17 +95,25 @@ class HttpServer {
     ::shutdown(listen_fd_, SHUT_RDWR);
     ::close(listen_fd_);
     if (accept_thread_.joinable()) accept_thread_.join();
-    // close live connections so their threads exit
+    // wake live connections so their threads exit, then join every thread
+    // BEFORE destroying its Conn record (threads only touch their own
+    // record, never the container, so join-then-destroy is race-free)
     {
       std::lock_guard<std::mutex> g(conn_mu_);
-      for (int fd : conn_fds_) ::shutdown(fd, SHUT_RDWR);
+      for (auto& c : conns_) ::shutdown(c->fd, SHUT_RDWR);
     }
-    for (auto& t : conn_threads_)
-      if (t.joinable()) t.join();
-    conn_threads_.clear();
+    for (auto& c : conns_)
+      if (c->th.joinable()) c->th.join();
+    conns_.clear();
   }
 
  private:
+  struct Conn {
+    int fd = -1;
+    std::atomic<bool> done{false};
+    std::thread th;
+  };
+
   void accept_loop() {
     while (running_) {
       int fd = ::accept(listen_fd_, nullptr, nullptr);
@@ -120,28 +129,37 @@ class HttpServer {
       timeval tv{300, 0};
       setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
       setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
+      auto conn = std::make_unique<Conn>();
+      conn->fd = fd;
+      Conn* cp = conn.get();
       {
         std::lock_guard<std::mutex> g(conn_mu_);
-        if (static_cast<int>(conn_fds_.size()) >= max_connections_) {
+        reap_finished_locked();
+        if (static_cast<int>(conns_.size()) >= max_connections_) {
           ::close(fd);
           continue;
         }
-        conn_fds_.push_back(fd);
-        reap_finished_locked();
-        conn_threads_.emplace_back([this, fd] { connection_loop(fd); });
+        conns_.push_back(std::move(conn));
+      }
+      cp->th = std::thread([this, cp] { connection_loop(cp->fd, cp); });
+    }
+  }
+
+  // Join and drop records of finished connections. A connection thread only
+  // ever touches ITS OWN Conn record (sets `done` last), so joining before
+  // destroying the record is race-free — no detach, no use-after-free.
+  void reap_finished_locked() {
+    for (auto it = conns_.begin(); it != conns_.end();) {
+      if ((*it)->done.load(std::memory_order_acquire)) {
+        if ((*it)->th.joinable()) (*it)->th.join();
+        it = conns_.erase(it);
+      } else {
+        ++it;
       }
     }
   }
 
-  void reap_finished_locked() {
-    // bound the thread vector: join threads whose fds are gone
-    if (conn_threads_.size() < 1024) return;
-    for (auto& t : conn_threads_)
-      if (t.joinable()) t.detach();
-    conn_threads_.clear();
-  }
-
-  void connection_loop(int fd) {
+  void connection_loop(int fd, Conn* self) {
     std::string buf;
     buf.reserve(8192);
     char chunk[16384];
@@ -238,9 +256,7 @@ class HttpServer {
     }
   done:
     ::close(fd);
-    std::lock_guard<std::mutex> g(conn_mu_);
-    conn_fds_.erase(std::remove(conn_fds_.begin(), conn_fds_.end(), fd),
-                    conn_fds_.end());
+    self->done.store(true, std::memory_order_release);  // LAST touch
   }
 
   std::string host_;
@@ -251,8 +267,7 @@ class HttpServer {
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
   std::mutex conn_mu_;
-  std::vector<int> conn_fds_;
-  std::vector<std::thread> conn_threads_;
+  std::vector<std::unique_ptr<Conn>> conns_;
 };
 
 }  // namespace egshttp
