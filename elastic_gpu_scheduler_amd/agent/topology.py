@@ -76,6 +76,66 @@ def parse_showtopohops(text: str) -> List[List[int]]:
     return matrix
 
 
+def parse_amd_smi_topology(payload: str) -> List[List[int]]:
+    """Parse `amd-smi topology --json`.
+
+    Common shape: a list of per-GPU objects with a "links" array:
+    [{"gpu": 0, "links": [{"gpu": 1, "link_type": "XGMI",
+                           "num_hops": 1}, ...]}, ...]
+    Accepts hop counts from "num_hops"/"hops"; non-XGMI links score as
+    routed (3). Exercised against canned output in tests.
+    """
+    try:
+        data = json.loads(payload)
+    except json.JSONDecodeError:
+        return []
+    if isinstance(data, dict):
+        data = data.get("topology", data.get("gpus", []))
+    if not isinstance(data, list) or not data:
+        return []
+    n = 0
+    entries = []
+    for entry in data:
+        if not isinstance(entry, dict) or "gpu" not in entry:
+            continue
+        entries.append(entry)
+        n = max(n, int(entry["gpu"]) + 1)
+        for link in entry.get("links", []) or []:
+            if isinstance(link, dict) and "gpu" in link:
+                n = max(n, int(link["gpu"]) + 1)
+    if not entries or n == 0:
+        return []
+    matrix = [[0 if i == j else 3 for j in range(n)] for i in range(n)]
+    for entry in entries:
+        i = int(entry["gpu"])
+        for link in entry.get("links", []) or []:
+            if not isinstance(link, dict) or "gpu" not in link:
+                continue
+            j = int(link["gpu"])
+            if i == j:
+                continue
+            ltype = str(link.get("link_type", link.get("type", ""))).upper()
+            hops = link.get("num_hops", link.get("hops", 1))
+            try:
+                hops = max(int(hops), 1)
+            except (TypeError, ValueError):
+                hops = 1
+            matrix[i][j] = hops if "XGMI" in ltype else 3
+    return matrix
+
+
+def _via_amd_smi() -> List[List[int]]:
+    exe = shutil.which("amd-smi")
+    if not exe:
+        return []
+    try:
+        out = subprocess.run([exe, "topology", "--json"], capture_output=True,
+                             text=True, timeout=30)
+        return parse_amd_smi_topology(out.stdout)
+    except (subprocess.SubprocessError, OSError):
+        return []
+
+
 def _via_rocm_smi() -> List[List[int]]:
     exe = shutil.which("rocm-smi")
     if not exe:
@@ -89,11 +149,12 @@ def _via_rocm_smi() -> List[List[int]]:
 
 
 def discover(n_cards: int, prefer: str = "auto") -> List[List[int]]:
-    sources = {"gpuprobe": _via_gpuprobe, "rocm-smi": _via_rocm_smi}
+    sources = {"gpuprobe": _via_gpuprobe, "amd-smi": _via_amd_smi,
+               "rocm-smi": _via_rocm_smi}
     if prefer in sources:
         m = sources[prefer]()
         return m if m else default_hive(n_cards)
-    for name in ("gpuprobe", "rocm-smi"):
+    for name in ("gpuprobe", "amd-smi", "rocm-smi"):
         try:
             m = sources[name]()
         except Exception:

@@ -149,3 +149,23 @@ def test_agent_publish_includes_allocatable(monkeypatch):
     node = client.get_node("gpu-node")
     assert node["status"]["allocatable"]["elasticgpu.io/gpu-core"] == "800"
     assert node["status"]["capacity"]["amd.com/gpu"] == "8"
+
+
+def test_parse_amd_smi_topology():
+    payload = json.dumps([
+        {"gpu": 0, "links": [
+            {"gpu": 1, "link_type": "XGMI", "num_hops": 1},
+            {"gpu": 2, "link_type": "PCIE", "num_hops": 2}]},
+        {"gpu": 1, "links": [
+            {"gpu": 0, "link_type": "XGMI", "num_hops": 1},
+            {"gpu": 2, "link_type": "XGMI", "num_hops": 2}]},
+        {"gpu": 2, "links": []},
+    ])
+    m = topo.parse_amd_smi_topology(payload)
+    assert m[0][1] == 1          # direct xGMI
+    assert m[0][2] == 3          # PCIe link counts as routed
+    assert m[1][2] == 2          # 2-hop xGMI
+    assert m[2][0] == 3          # unreported direction defaults to routed
+    assert [m[i][i] for i in range(3)] == [0, 0, 0]
+    assert topo.parse_amd_smi_topology("junk") == []
+    assert topo.parse_amd_smi_topology("[]") == []
