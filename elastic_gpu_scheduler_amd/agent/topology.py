@@ -13,6 +13,7 @@ Sources, in order:
 """
 from __future__ import annotations
 
+import json
 import logging
 import re
 import shutil
