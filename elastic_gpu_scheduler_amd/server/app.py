@@ -270,7 +270,7 @@ class _BadRequest(Exception):
 def _parse_json(body: bytes) -> Dict[str, Any]:
     try:
         out = json.loads(body or b"{}")
-    except json.JSONDecodeError as exc:
+    except (json.JSONDecodeError, UnicodeDecodeError) as exc:
         raise _BadRequest(f"malformed JSON: {exc}") from exc
     if not isinstance(out, dict):
         raise _BadRequest("JSON body must be an object")

@@ -172,3 +172,10 @@ def test_debug_profile_collapsed_stacks(cluster, extender):
     stack, _, count = line.rpartition(" ")
     assert int(count) >= 1
     assert ":" in stack
+
+
+def test_non_utf8_body_is_400(extender):
+    r = extender.request("POST", "/scheduler/filter", content=b"\xe1\x97\x9f")
+    assert r.status_code == 400
+    r = extender.request("POST", "/scheduler/bind", content=b"\xff\xfe")
+    assert r.status_code == 400
