@@ -142,11 +142,20 @@ class MiniHttpClient:
         head, _, rest = self.buf.partition(b"\r\n\r\n")
         lines = head.split(b"\r\n")
         status = int(lines[0].split()[1])
-        clen = 0
+        clen = None
+        chunked = False
         for line in lines[1:]:
             k, _, v = line.partition(b":")
-            if k.strip().lower() == b"content-length":
+            key = k.strip().lower()
+            if key == b"content-length":
                 clen = int(v.strip())
+            elif key == b"transfer-encoding" and b"chunked" in v.lower():
+                chunked = True
+        if chunked:
+            body, rest = self._read_chunked(rest)
+            self.buf = rest
+            return status, json.loads(body or b"{}")
+        clen = clen or 0
         while len(rest) < clen:
             chunk = self.sock.recv(65536)
             if not chunk:
@@ -154,6 +163,26 @@ class MiniHttpClient:
             rest += chunk
         self.buf = rest[clen:]
         return status, json.loads(rest[:clen] or b"{}")
+
+    def _read_chunked(self, rest: bytes) -> "tuple[bytes, bytes]":
+        body = b""
+        while True:
+            while b"\r\n" not in rest:
+                chunk = self.sock.recv(65536)
+                if not chunk:
+                    raise ConnectionError("server closed connection")
+                rest += chunk
+            size_line, _, rest = rest.partition(b"\r\n")
+            size = int(size_line.split(b";")[0], 16)
+            while len(rest) < size + 2:
+                chunk = self.sock.recv(65536)
+                if not chunk:
+                    raise ConnectionError("server closed connection")
+                rest += chunk
+            body += rest[:size]
+            rest = rest[size + 2:]  # skip trailing CRLF
+            if size == 0:
+                return body, rest
 
 
 

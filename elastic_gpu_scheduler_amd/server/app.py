@@ -106,7 +106,9 @@ class ExtenderApp:
                 break
         status, ctype, out = self.handle(method, path, body)
         await send({"type": "http.response.start", "status": status,
-                    "headers": [(b"content-type", ctype.encode())]})
+                    "headers": [(b"content-type", ctype.encode()),
+                                (b"content-length",
+                                 str(len(out)).encode())]})
         await send({"type": "http.response.body", "body": out})
 
     # ---- handlers --------------------------------------------------------
