@@ -41,6 +41,7 @@ class Controller:
         self._pending: Dict[str, Dict[str, Any]] = {}  # key -> last seen pod
         self._retries: Dict[str, int] = {}
         self.max_retries = 5
+        self._node_rv: Dict[str, Any] = {}  # node -> last seen resourceVersion
         self._pending_mu = threading.Lock()
         self._stop = threading.Event()
         self._threads: list[threading.Thread] = []
@@ -157,9 +158,34 @@ class Controller:
                 continue
             live.add(obj.pod_uid(pod))
             self._sync_pod(pod)
+        schedulers = {id(s): s for s in self.registry.schedulers.values()}
         # Evict accounting for pods that vanished without a DELETE event.
-        for sch in {id(s): s for s in self.registry.schedulers.values()}.values():
+        for sch in schedulers.values():
             for node in sch.state.node_names():
                 for uid in sch.state.node_pods(node):
                     if uid not in live:
                         sch.state.forget_pod(uid)
+        self._resync_nodes(schedulers.values())
+
+    def _resync_nodes(self, schedulers) -> None:
+        """Invalidate node caches whose objects changed (the agent
+        republishing inventory/topology bumps resourceVersion) or vanished;
+        the next use lazily refills and replays assumed pods."""
+        try:
+            nodes = self.client.list_nodes()
+        except Exception:
+            log.debug("node relist failed", exc_info=True)
+            return
+        current = {n["metadata"]["name"]: n["metadata"].get("resourceVersion")
+                   for n in nodes if n.get("metadata", {}).get("name")}
+        for sch in schedulers:
+            for name in sch.state.node_names():
+                if name not in current:
+                    log.info("node %s deleted; evicting from cache", name)
+                    sch.invalidate_node(name)
+                elif self._node_rv.get(name) is not None and \
+                        current[name] != self._node_rv[name]:
+                    log.info("node %s changed (rv %s -> %s); refreshing cache",
+                             name, self._node_rv[name], current[name])
+                    sch.invalidate_node(name)
+        self._node_rv = current

@@ -153,3 +153,34 @@ def test_sync_failure_retries_with_backoff():
         assert calls["n"] >= 3
     finally:
         ctrl.stop()
+
+
+def test_resync_refreshes_changed_node_inventory():
+    """When the agent republishes a node's inventory (resourceVersion
+    bumps), the resync evicts the cached allocator; the next use re-reads
+    the fresh inventory and replays assumed pods."""
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=25, memory=GiB))
+        sch.assume(["n1"], pod)
+        sch.bind("n1", pod)
+        ctrl.resync_once()  # records the baseline resourceVersion
+        assert len(sch.state.node_devices("n1")) == 8
+
+        # agent publishes a shrunk inventory (e.g. a sick card excluded)
+        import json as _json
+
+        cards = [{"index": i, "memory_bytes": 288 * GiB, "core": 100}
+                 for i in range(4)]
+        client.patch_node_annotations("n1", {
+            "elasticgpu.io/gpu-inventory": _json.dumps({"cards": cards})})
+        ctrl.resync_once()  # detects the change and evicts
+        ok, _ = sch.assume(["n1"], client.create_pod(make_pod("q", core=10)))
+        assert ok == ["n1"]
+        devs = sch.state.node_devices("n1")
+        assert len(devs) == 4  # fresh inventory picked up
+        # the bound pod's accounting survived the refresh (replayed)
+        assert any(d.core_avail == 75 for d in devs)
+    finally:
+        ctrl.stop()
