@@ -216,6 +216,9 @@ class ExtenderApp:
                                          self.registry.schedulers.values()})}, None
 
     def metrics(self, body: bytes):
+        # refresh point-in-time gauges (the native fast path doesn't touch
+        # Python metrics per request)
+        metrics.NODES_CACHED.set(len(self.registry.default.state.node_names()))
         out = metrics.render()
         if self.native_server is not None:
             stats = self.native_server.stats()
