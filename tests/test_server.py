@@ -179,3 +179,48 @@ def test_non_utf8_body_is_400(extender):
     assert r.status_code == 400
     r = extender.request("POST", "/scheduler/bind", content=b"\xff\xfe")
     assert r.status_code == 400
+
+
+def test_qgpu_mode_end_to_end(fake_client):
+    """qgpu mode: only qgpu resources route; full cycle over the app."""
+    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+    from elastic_gpu_scheduler_amd.server.app import make_app
+    from tests.conftest import ExtenderClient, make_node
+
+    fake_client.add_node(make_node("node-a"))
+    registry = SchedulerRegistry(fake_client, mode="qgpu")
+    ext = ExtenderClient(make_app(registry))
+
+    qpod = fake_client.create_pod({
+        "metadata": {"name": "q", "namespace": "default", "uid": "qu"},
+        "spec": {"containers": [{"name": "c", "resources": {"requests": {
+            "elasticgpu.io/qgpu-core": "40",
+            "elasticgpu.io/qgpu-memory": "32Gi"}}}]},
+        "status": {"phase": "Pending"}})
+    r = ext.filter(qpod, ["node-a"])
+    assert r.json()["nodenames"] == ["node-a"]
+    assert ext.bind(qpod, "node-a").status_code == 200
+    bound = fake_client.get_pod("default", "q")
+    assert bound["spec"]["nodeName"] == "node-a"
+    assert bound["metadata"]["annotations"]["elasticgpu.io/container-c"] == "0"
+
+    # gpushare pods are NOT managed in qgpu mode: pass-through
+    gpod = fake_client.create_pod(make_pod("g", core=30))
+    r = ext.filter(gpod, ["node-a"])
+    assert r.json()["nodenames"] == ["node-a"]  # untouched
+
+
+def test_pgpu_mode_end_to_end(fake_client):
+    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+    from elastic_gpu_scheduler_amd.server.app import make_app
+    from tests.conftest import ExtenderClient, make_node
+
+    fake_client.add_node(make_node("node-a", cards=2))
+    registry = SchedulerRegistry(fake_client, mode="pgpu")
+    ext = ExtenderClient(make_app(registry))
+    pod = fake_client.create_pod(make_pod("p", per_container=[{"pgpu": 2}]))
+    r = ext.filter(pod, ["node-a"])
+    assert r.json()["nodenames"] == ["node-a"]
+    assert ext.bind(pod, "node-a").status_code == 200
+    bound = fake_client.get_pod("default", "p")
+    assert bound["metadata"]["annotations"]["elasticgpu.io/container-c0"] == "0,1"
