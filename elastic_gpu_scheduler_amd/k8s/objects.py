@@ -201,7 +201,12 @@ def node_devices(node: Node, bare_unit: str = "auto") -> List[Any]:
         try:
             inv = json.loads(inv_raw)
             devices = []
-            for card in inv.get("cards", []):
+            # device list position == physical card index: sort defensively
+            # (an agent publishing cards out of order must not skew the
+            # annotation index -> physical card mapping)
+            cards = sorted(inv.get("cards", []),
+                           key=lambda c: int(c.get("index", 0)))
+            for card in cards:
                 mem = int(card.get("memory_bytes", t.MI355X_MEMORY_BYTES))
                 devices.append(core.Device(
                     core_total=int(card.get("core", t.GPU_CORE_EACH_CARD)),

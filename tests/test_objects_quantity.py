@@ -151,3 +151,15 @@ class TestNodeInventory:
             "elasticgpu.io/xgmi-topology": "also not"})
         assert len(obj.node_devices(node)) == 2  # falls back to allocatable
         assert obj.node_topology(node) == []
+
+    def test_agent_inventory_out_of_order_cards(self):
+        import json
+
+        node = make_node("n", annotations={
+            "elasticgpu.io/gpu-inventory": json.dumps({"cards": [
+                {"index": 1, "core": 100, "memory_bytes": 144 * GiB},
+                {"index": 0, "core": 100, "memory_bytes": 288 * GiB},
+            ]})})
+        devs = obj.node_devices(node)
+        assert devs[0].mem_total == 288 * GiB  # position == physical index
+        assert devs[1].mem_total == 144 * GiB
