@@ -160,6 +160,19 @@ PYBIND11_MODULE(_core, m) {
              py::gil_scoped_release rel;
              return alloc->assumed_count();
            })
+      .def("node_pod_placements",
+           [](ClusterState& cs, const std::string& name) {
+             auto alloc = cs.get(name);
+             if (!alloc) throw std::runtime_error("unknown node " + name);
+             std::vector<std::pair<std::string, std::vector<std::vector<int>>>> out;
+             {
+               py::gil_scoped_release rel;
+               out = alloc->pod_placements();
+             }
+             py::dict d;
+             for (auto& [uid, cards] : out) d[py::str(uid)] = cards;
+             return d;
+           })
       .def("node_pods", [](ClusterState& cs, const std::string& name) {
         auto alloc = cs.get(name);
         if (!alloc) throw std::runtime_error("unknown node " + name);

@@ -173,6 +173,15 @@ class NodeAllocator {
     return static_cast<int>(assumed_.size());
   }
 
+  std::vector<std::pair<std::string, std::vector<std::vector<int>>>>
+  pod_placements() {
+    std::lock_guard<std::mutex> g(mu_);
+    std::vector<std::pair<std::string, std::vector<std::vector<int>>>> out;
+    out.reserve(pods_.size());
+    for (const auto& [uid, opt] : pods_) out.emplace_back(uid, opt.allocated);
+    return out;
+  }
+
   std::vector<std::string> pod_uids() {
     std::lock_guard<std::mutex> g(mu_);
     std::vector<std::string> out;

@@ -290,7 +290,7 @@ class GPUUnitScheduler:
                     "memory_available": d.mem_avail,
                     "memory_total": d.mem_total,
                 } for d in devices],
-                "pods": self.state.node_pods(name),
+                "pods": self.state.node_pod_placements(name),
             }
         return {"name": self.name, "policy": self.policy, "nodes": nodes}
 

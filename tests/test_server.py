@@ -124,6 +124,9 @@ def test_status_version_metrics_healthz(cluster, extender):
     gpus = st["gpushare"]["nodes"]["node-a"]["gpus"]
     assert len(gpus) == 8
     assert any(g["core_available"] == 70 for g in gpus)
+    pods = st["gpushare"]["nodes"]["node-a"]["pods"]
+    assert list(pods.values()) == [[[0]]] or \
+        any(len(v) == 1 and len(v[0]) == 1 for v in pods.values())
 
     assert extender.request("GET", "/version").json()["target"].startswith("MI355X")
     m = extender.request("GET", "/metrics")
