@@ -88,7 +88,7 @@ PYBIND11_MODULE(_core, m) {
       "search_placement",
       [](const std::vector<Device>& devices, const GPURequest& req,
          const std::string& policy, uint64_t seed,
-         std::vector<std::vector<int>> topo_hops) {
+         std::vector<std::vector<int>> topo_hops, bool distinct) {
         auto rater = make_rater(policy, seed);
         Topology topo(std::move(topo_hops));
         RateContext ctx;
@@ -98,12 +98,13 @@ PYBIND11_MODULE(_core, m) {
         SearchResult res;
         {
           py::gil_scoped_release rel;
-          res = search_placement(devices, req, *rater, ctx);
+          res = search_placement(devices, req, *rater, ctx, distinct);
         }
         return py::make_tuple(res.feasible, res.option, res.leaves_evaluated);
       },
       py::arg("devices"), py::arg("request"), py::arg("policy") = "binpack",
-      py::arg("seed") = 0, py::arg("topology") = std::vector<std::vector<int>>{});
+      py::arg("seed") = 0, py::arg("topology") = std::vector<std::vector<int>>{},
+      py::arg("distinct") = false);
 
   bind_native_server(m);
 
