@@ -129,3 +129,28 @@ def test_whole_card_picks_topology_best_not_first_free():
     feasible, opt, _ = core.search_placement(devs, units, "binpack", 0, hops)
     assert feasible
     assert sorted(opt.allocated[0]) == [2, 3]
+
+
+def test_combined_constraints_hive_plus_spread():
+    """Whole-card hive locality + spread-containers + fractional mix, all at
+    once, under the leaf budget (the combination the demo's hardest pod
+    exercises)."""
+    hops = [[0 if i == j else (1 if (i < 8) == (j < 8) else 3)
+             for j in range(16)] for i in range(16)]
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(16)]
+    units = [core.GPUUnit(6, 0, 0), core.GPUUnit(0, 40, 8 * GiB),
+             core.GPUUnit(0, 40, 8 * GiB)]
+    feasible, opt, leaves = core.search_placement(devs, units, "binpack", 0,
+                                                  hops, True)
+    assert feasible and leaves <= 4096
+    assert len({i < 8 for i in opt.allocated[0]}) == 1  # in one hive
+    all_cards = [c for a in opt.allocated for c in a]
+    assert len(all_cards) == len(set(all_cards))  # spread honoured
+
+
+def test_huge_subset_space_stays_bounded():
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(20)]
+    feasible, opt, leaves = core.search_placement(
+        devs, [core.GPUUnit(10, 0, 0)], "binpack", 0, [])
+    assert feasible and len(set(opt.allocated[0])) == 10
+    assert leaves <= 4096  # C(20,10)=184,756 raw combinations
