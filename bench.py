@@ -456,9 +456,29 @@ def main():
     if use_gpu:
         torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
     if distributed:
+        import contextlib
         import torch.distributed as dist
 
-        dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+        @contextlib.contextmanager
+        def _quiet_stdout():
+            # gloo prints its rendezvous banner straight to fd 1, which
+            # would pollute the single-JSON-line stdout contract
+            import sys as _sys
+
+            _sys.stdout.flush()
+            saved = os.dup(1)
+            devnull = os.open(os.devnull, os.O_WRONLY)
+            os.dup2(devnull, 1)
+            try:
+                yield
+            finally:
+                _sys.stdout.flush()
+                os.dup2(saved, 1)
+                os.close(saved)
+                os.close(devnull)
+
+        with _quiet_stdout():
+            dist.init_process_group(backend="nccl" if use_gpu else "gloo")
 
     device_index = local_rank % max(
         torch.cuda.device_count(), 1) if use_gpu else 0
