@@ -107,6 +107,12 @@ class ParseError : public std::runtime_error {
 
 namespace detail {
 
+// Recursion guard: the parser descends once per nesting level, so without a
+// bound a body of 100k '[' characters would overflow the C stack (found by
+// adversarial review; Python's json raises RecursionError at a similar
+// depth). Kubernetes objects nest ~10 deep.
+constexpr int kMaxDepth = 256;
+
 class Parser {
  public:
   Parser(const char* data, size_t len) : p_(data), end_(data + len) {}
@@ -139,6 +145,13 @@ class Parser {
   }
 
   Value parse_value() {
+    struct DepthGuard {
+      int& d;
+      explicit DepthGuard(int& d_) : d(d_) {
+        if (++d > kMaxDepth) throw ParseError("nesting too deep");
+      }
+      ~DepthGuard() { --d; }
+    } guard(depth_);
     switch (peek()) {
       case '{': return parse_object();
       case '[': return parse_array();
@@ -289,6 +302,7 @@ class Parser {
 
   const char* p_;
   const char* end_;
+  int depth_ = 0;
 };
 
 }  // namespace detail

@@ -277,6 +277,8 @@ def _parse_json(body: bytes) -> Dict[str, Any]:
         out = json.loads(body or b"{}")
     except (json.JSONDecodeError, UnicodeDecodeError) as exc:
         raise _BadRequest(f"malformed JSON: {exc}") from exc
+    except RecursionError as exc:  # nesting bomb (the C++ codec caps at 256)
+        raise _BadRequest("JSON nesting too deep") from exc
     if not isinstance(out, dict):
         raise _BadRequest("JSON body must be an object")
     return out

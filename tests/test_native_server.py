@@ -267,3 +267,18 @@ def test_native_connection_churn_and_partial_requests(native):
         r = c.post("/scheduler/filter",
                    json={"pod": pod, "nodenames": ["node-a"]})
         assert r.status_code == 200
+
+
+def test_nesting_bomb_rejected_not_crashed(native):
+    """A deeply-nested JSON body must produce a 4xx, crash neither the C++
+    parser (stack-depth cap) nor the Python fallback (RecursionError ->
+    400)."""
+    _, _, fe = native
+    with _client(fe) as c:
+        r = c.post("/scheduler/filter", content=b"[" * 500000,
+                   headers={"content-type": "application/json"})
+        assert r.status_code == 400
+        assert c.get("/healthz").status_code == 200
+    assert core.json_roundtrip("[" * 200 + "1" + "]" * 200)
+    with pytest.raises(RuntimeError):
+        core.json_roundtrip("[" * 1000 + "1" + "]" * 1000)
