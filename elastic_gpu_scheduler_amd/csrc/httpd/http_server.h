@@ -170,7 +170,7 @@ class HttpServer {
         ssize_t n = ::recv(fd, chunk, sizeof(chunk), 0);
         if (n <= 0) goto done;
         buf.append(chunk, n);
-        if (buf.size() > (16u << 20)) goto done;  // 16 MiB header bound
+        if (buf.size() > (1u << 20)) goto done;  // 1 MiB header bound
       }
       {
         Request req;
@@ -206,7 +206,10 @@ class HttpServer {
               } catch (const std::exception&) {
                 goto done;
               }
-              if (content_length > (64u << 20)) goto done;  // 64 MiB bound
+              // 8 MiB bound: a pod object is capped ~1.5 MiB by etcd and
+              // ExtenderArgs adds only node names; larger bodies are abuse
+              // (and 512 conns x large maps would pressure memory)
+              if (content_length > (8u << 20)) goto done;
             } else if (key == "connection") {
               for (auto& c : value) c = static_cast<char>(tolower(c));
               keep_alive = value != "close";

@@ -184,10 +184,10 @@ def test_native_http_edge_cases(native):
     assert s.recv(1024) == b""  # closed
     s.close()
 
-    # oversized content-length refused (connection closed)
+    # oversized content-length refused (connection closed; cap is 8 MiB)
     s = _socket.create_connection(("127.0.0.1", fe.port))
     s.sendall(b"POST /scheduler/filter HTTP/1.1\r\nhost: t\r\n"
-              b"content-length: 99999999999\r\n\r\n")
+              b"content-length: 99999999\r\n\r\n")
     s.settimeout(5)
     assert s.recv(1024) == b""
     s.close()
