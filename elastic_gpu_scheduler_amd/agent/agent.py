@@ -87,6 +87,34 @@ class NodeAgent:
             pass  # client without status-subresource support
         return ann
 
+    def publish_with_health(self, mib: int = 256, iters: int = 5
+                            ) -> Dict[str, Any]:
+        """Publish inventory with unhealthy cards EXCLUDED (HBM bandwidth
+        below threshold): the scheduler stops placing onto sick cards at the
+        next node-cache refresh. Returns {"published": ann, "sick": [...]}.
+        """
+        if self.client is None:
+            raise RuntimeError("publish_with_health needs a KubeClient")
+        report = self.health_check(mib=mib, iters=iters)
+        sick = [r["index"] for r in report if not r["healthy"]]
+        ann = self.annotations()
+        inv = json.loads(ann[t.ANNOTATION_NODE_INVENTORY])
+        inv["cards"] = [c for c in inv["cards"] if c["index"] not in sick]
+        ann[t.ANNOTATION_NODE_INVENTORY] = json.dumps(inv)
+        self.client.patch_node_annotations(self.node_name, ann)
+        try:
+            alloc = {
+                t.RESOURCE_GPU_CORE: str(sum(c.get("core", 100)
+                                             for c in inv["cards"])),
+                t.RESOURCE_GPU_MEMORY: str(sum(int(c.get("memory_bytes", 0))
+                                               for c in inv["cards"])),
+                t.RESOURCE_AMD_GPU: str(len(inv["cards"])),
+            }
+            self.client.patch_node_allocatable(self.node_name, alloc)
+        except NotImplementedError:
+            pass
+        return {"published": ann, "sick": sick}
+
     def node_object(self) -> Dict[str, Any]:
         """A complete Node object for offline/bench use (fake apiserver)."""
         return {

@@ -67,17 +67,10 @@ def main(argv=None) -> int:
     while True:
         try:
             if args.health_check:
-                report = agent.health_check()
-                sick = [r["index"] for r in report if not r["healthy"]]
-                if sick:
+                out = agent.publish_with_health()
+                if out["sick"]:
                     log.warning("unhealthy cards (excluded from inventory): %s",
-                                sick)
-                ann = agent.annotations()
-                inv = json.loads(ann["elasticgpu.io/gpu-inventory"])
-                inv["cards"] = [c for c in inv["cards"]
-                                if c["index"] not in sick]
-                ann["elasticgpu.io/gpu-inventory"] = json.dumps(inv)
-                client.patch_node_annotations(args.node, ann)
+                                out["sick"])
             else:
                 agent.publish()
             log.info("published inventory for %s", args.node)

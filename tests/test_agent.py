@@ -169,3 +169,29 @@ def test_parse_amd_smi_topology():
     assert [m[i][i] for i in range(3)] == [0, 0, 0]
     assert topo.parse_amd_smi_topology("junk") == []
     assert topo.parse_amd_smi_topology("[]") == []
+
+
+def test_publish_with_health_excludes_sick_cards(monkeypatch):
+    """A card failing the HBM health gate disappears from the published
+    inventory and allocatable, so the scheduler stops placing onto it."""
+    cards = [{"index": i, "memory_bytes": 288 * GiB, "core": 100}
+             for i in range(4)]
+    monkeypatch.setattr(inv, "discover", lambda prefer="auto": cards)
+    monkeypatch.setattr(topo, "discover",
+                        lambda n, prefer="auto": topo.default_hive(n))
+    client = FakeKubeClient()
+    client.add_node({"metadata": {"name": "gpu-node"}, "status": {}})
+    agent = NodeAgent("gpu-node", client)
+    monkeypatch.setattr(NodeAgent, "health_check",
+                        lambda self, mib=256, iters=5: [
+                            {"index": 0, "hbm_gbps": 6100.0, "healthy": True},
+                            {"index": 1, "hbm_gbps": 240.0, "healthy": False},
+                            {"index": 2, "hbm_gbps": 6050.0, "healthy": True},
+                            {"index": 3, "hbm_gbps": 6200.0, "healthy": True}])
+    out = agent.publish_with_health()
+    assert out["sick"] == [1]
+    node = client.get_node("gpu-node")
+    devs = obj.node_devices(node)
+    assert len(devs) == 3
+    assert node["status"]["allocatable"]["elasticgpu.io/gpu-core"] == "300"
+    assert node["status"]["allocatable"]["amd.com/gpu"] == "3"
