@@ -29,6 +29,13 @@ except ImportError as exc:  # pragma: no cover - build failure is fatal
         "Build it in-tree with `python build_native.py` (repo root)."
     ) from exc
 
+# Push the bare-number "auto" GiB/bytes threshold into the C++ fast path so
+# there is ONE definition (utils/quantity.py) — a drift between the two would
+# split memory-quantity semantics per request path.
+from elastic_gpu_scheduler_amd.utils.quantity import BARE_AUTO_GIB_THRESHOLD as _thr
+
+core.set_bare_auto_gib_threshold(_thr)
+
 _gpuprobe = None
 _gpuprobe_err: Exception | None = None
 _gpuprobe_tried = False

@@ -99,7 +99,18 @@ class NodeAgent:
         sick = [r["index"] for r in report if not r["healthy"]]
         ann = self.annotations()
         inv = json.loads(ann[t.ANNOTATION_NODE_INVENTORY])
-        inv["cards"] = [c for c in inv["cards"] if c["index"] not in sick]
+        # Sick cards stay in the list as zero-capacity placeholders so list
+        # position keeps equalling the PHYSICAL card index — dropping an
+        # entry would shift every later card's index and the scheduler's
+        # annotations (consumed as physical indexes by the device plugin)
+        # would map pods onto the wrong card, including the sick one.
+        # Zero-capacity devices are never schedulable (csrc/core/types.h
+        # Device::schedulable).
+        inv["cards"] = [
+            ({**c, "core": 0, "memory_bytes": 0, "sick": True}
+             if c["index"] in sick else c)
+            for c in inv["cards"]
+        ]
         ann[t.ANNOTATION_NODE_INVENTORY] = json.dumps(inv)
         self.client.patch_node_annotations(self.node_name, ann)
         try:
@@ -108,7 +119,8 @@ class NodeAgent:
                                              for c in inv["cards"])),
                 t.RESOURCE_GPU_MEMORY: str(sum(int(c.get("memory_bytes", 0))
                                                for c in inv["cards"])),
-                t.RESOURCE_AMD_GPU: str(len(inv["cards"])),
+                t.RESOURCE_AMD_GPU: str(sum(1 for c in inv["cards"]
+                                            if not c.get("sick"))),
             }
             self.client.patch_node_allocatable(self.node_name, alloc)
         except NotImplementedError:

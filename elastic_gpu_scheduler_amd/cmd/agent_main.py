@@ -10,8 +10,9 @@ GPU nodes (deploy/elastic-gpu-agent-amd.yaml):
 Publishes:
   elasticgpu.io/gpu-inventory   per-card {core, memory_bytes, name, arch}
   elasticgpu.io/xgmi-topology   hop matrix for locality-aware placement
-and optionally runs the HBM health probe each cycle, dropping sick cards
-from the published inventory so the scheduler stops placing pods on them.
+and optionally runs the HBM health probe each cycle, publishing sick cards
+as zero-capacity placeholders (list position always equals the physical card
+index) so the scheduler stops placing pods on them.
 """
 from __future__ import annotations
 
@@ -33,7 +34,8 @@ def build_parser() -> argparse.ArgumentParser:
                    help="run the HBM bandwidth probe each cycle and exclude "
                         "unhealthy cards from the inventory")
     p.add_argument("--source", default="auto",
-                   choices=("auto", "gpuprobe", "amdsmi", "rocm-smi", "torch"))
+                   choices=("auto", "gpuprobe", "amdsmi", "amd-smi",
+                            "rocm-smi", "torch"))
     p.add_argument("--dry-run", action="store_true",
                    help="print the annotations instead of patching the node")
     p.add_argument("--log-level", default="info")
@@ -69,7 +71,7 @@ def main(argv=None) -> int:
             if args.health_check:
                 out = agent.publish_with_health()
                 if out["sick"]:
-                    log.warning("unhealthy cards (excluded from inventory): %s",
+                    log.warning("unhealthy cards (published zero-capacity): %s",
                                 out["sick"])
             else:
                 agent.publish()

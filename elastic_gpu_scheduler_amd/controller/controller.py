@@ -42,6 +42,7 @@ class Controller:
         self._retries: Dict[str, int] = {}
         self.max_retries = 5
         self._node_rv: Dict[str, Any] = {}  # node -> last seen resourceVersion
+        self._missing_once: set = set()  # uids absent from the last relist
         self._pending_mu = threading.Lock()
         self._stop = threading.Event()
         self._threads: list[threading.Thread] = []
@@ -159,12 +160,25 @@ class Controller:
             live.add(obj.pod_uid(pod))
             self._sync_pod(pod)
         schedulers = {id(s): s for s in self.registry.schedulers.values()}
-        # Evict accounting for pods that vanished without a DELETE event.
+        # Evict accounting for pods that vanished without a DELETE event —
+        # but only after they have been missing from TWO consecutive relists.
+        # The pod list is a snapshot taken before the sweep: a pod bound
+        # between list_pods() and the sweep would otherwise be wrongly
+        # forgotten, transiently over-freeing its cards until the next event
+        # re-adds it (another pod could double-book them in that window).
+        missing_now = set()
         for sch in schedulers.values():
             for node in sch.state.node_names():
                 for uid in sch.state.node_pods(node):
-                    if uid not in live:
+                    if uid in live:
+                        continue
+                    if uid in self._missing_once:
+                        log.info("pod uid %s missing from two consecutive "
+                                 "relists; forgetting", uid)
                         sch.state.forget_pod(uid)
+                    else:
+                        missing_now.add(uid)
+        self._missing_once = missing_now
         self._resync_nodes(schedulers.values())
 
     def _resync_nodes(self, schedulers) -> None:

@@ -45,6 +45,7 @@ PYBIND11_MODULE(_core, m) {
       .def_readwrite("mem_total", &Device::mem_total)
       .def_readwrite("mem_avail", &Device::mem_avail)
       .def("whole_free", &Device::whole_free)
+      .def("schedulable", &Device::schedulable)
       .def("__repr__", [](const Device& d) {
         return "Device(core " + std::to_string(d.core_avail) + "/" +
                std::to_string(d.core_total) + ", mem " + std::to_string(d.mem_avail) +

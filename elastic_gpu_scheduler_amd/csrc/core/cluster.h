@@ -56,16 +56,18 @@ class ThreadPool {
       return;
     }
     std::atomic<int> next{0};
-    std::atomic<int> done{0};
+    int done = 0;  // guarded by done_mu
     std::mutex done_mu;
     std::condition_variable done_cv;
     auto task = [&, n] {
       int i;
       while ((i = next.fetch_add(1)) < n) fn(i);
-      if (done.fetch_add(1) + 1 == size()) {
-        std::lock_guard<std::mutex> g(done_mu);
-        done_cv.notify_one();
-      }
+      // The final increment happens UNDER done_mu: if it were a bare atomic
+      // the waiting caller could observe done==size() on a spurious wakeup,
+      // return, and destroy done_mu/done_cv while this worker is about to
+      // lock them (use-after-destroy race).
+      std::lock_guard<std::mutex> g(done_mu);
+      if (++done == size()) done_cv.notify_one();
     };
     {
       std::lock_guard<std::mutex> g(mu_);
@@ -73,7 +75,7 @@ class ThreadPool {
     }
     cv_.notify_all();
     std::unique_lock<std::mutex> g(done_mu);
-    done_cv.wait(g, [&] { return done.load() == size(); });
+    done_cv.wait(g, [&] { return done == size(); });
   }
 
  private:

@@ -113,4 +113,14 @@ void bind_native_server(py::module_& m) {
   m.def("json_roundtrip", [](const std::string& s) {
     return egsjson::dump(egsjson::parse(s));
   });
+
+  // Bare-number "auto" threshold: Python utils/quantity.py is the single
+  // source of truth; package import pushes it here so both request paths
+  // (C++ fast path, Python fallback) always agree.
+  m.def("set_bare_auto_gib_threshold", [](int64_t v) {
+    bare_auto_gib_threshold().store(v, std::memory_order_relaxed);
+  });
+  m.def("get_bare_auto_gib_threshold", []() {
+    return bare_auto_gib_threshold().load(std::memory_order_relaxed);
+  });
 }

@@ -33,6 +33,11 @@ namespace egs {
 // (the first feasible leaf is always reached if one exists in the explored
 // prefix; the greedy order explores plausible placements first).
 constexpr int kMaxLeafEvals = 4096;
+// Upper bound on TOTAL dfs node visits (complete assignments AND dead-end
+// partial paths). kMaxLeafEvals alone only counts complete assignments: a
+// multi-container pod whose last container never fits would explore up to
+// cards^containers dead ends with zero leaves, stalling the node mutex.
+constexpr int kMaxDfsVisits = 65536;
 // Upper bound on whole-card k-subsets enumerated per container.
 constexpr int kMaxWholeCardCandidates = 128;
 
@@ -90,6 +95,7 @@ struct DfsState {
   GPUOption best;
   bool found = false;
   int leaves = 0;
+  int visits = 0;  // every dfs() entry, including dead ends
   // distinct_containers: no card may serve two containers of this pod
   // (elasticgpu.io/spread-containers annotation)
   bool distinct_containers = false;
@@ -126,7 +132,7 @@ inline void try_candidate(DfsState& st, size_t c, const std::vector<int>& cards,
 }
 
 inline void dfs(DfsState& st, size_t c) {
-  if (st.leaves >= kMaxLeafEvals) return;
+  if (st.leaves >= kMaxLeafEvals || ++st.visits > kMaxDfsVisits) return;
   const GPURequest& req = *st.req;
   if (c == req.size()) {
     ++st.leaves;
@@ -187,6 +193,32 @@ inline SearchResult search_placement(const std::vector<Device>& devices,
                                      const GPURequest& req, const Rater& rater,
                                      const RateContext& ctx,
                                      bool distinct_containers = false) {
+  // Fail fast when some container can never fit even on a completely FREE
+  // node: without this the dfs explores every placement of the earlier
+  // containers before discovering the doomed one (the kMaxDfsVisits cap
+  // bounds that walk; this check removes it entirely for the common case of
+  // an oversized container).
+  {
+    int schedulable = 0;
+    for (const auto& d : devices)
+      if (d.schedulable()) ++schedulable;
+    for (const auto& u : req) {
+      if (!u.needs_gpu()) continue;
+      if (u.whole_cards()) {
+        if (u.gpu_count > schedulable) return SearchResult{};
+      } else {
+        bool fits_somewhere = false;
+        for (const auto& d : devices) {
+          if (d.core_total >= u.core && d.mem_total >= u.memory &&
+              d.schedulable()) {
+            fits_somewhere = true;
+            break;
+          }
+        }
+        if (!fits_somewhere) return SearchResult{};
+      }
+    }
+  }
   search_detail::DfsState st;
   st.devices = devices;
   st.req = &req;

@@ -32,11 +32,17 @@ struct Device {
   int64_t mem_total = kMI355XMemoryBytes;
   int64_t mem_avail = kMI355XMemoryBytes;
 
+  // Zero-capacity devices are placeholders the agent publishes for sick
+  // cards (HBM health gate) so list position keeps equalling the physical
+  // card index; they must never look schedulable — without the core_total
+  // guard a 0/0 card is "whole free" (0 == 0) and a whole-card request
+  // would land on the sick card.
+  bool schedulable() const { return core_total > 0; }
   bool whole_free() const {
-    return core_avail == core_total && mem_avail == mem_total;
+    return schedulable() && core_avail == core_total && mem_avail == mem_total;
   }
   bool can_fit(int core, int64_t mem) const {
-    return core_avail >= core && mem_avail >= mem;
+    return schedulable() && core_avail >= core && mem_avail >= mem;
   }
 };
 

@@ -90,6 +90,16 @@ inline bool parse_quantity_ll(const std::string& s, int64_t* out,
 
 enum class BareUnit { Auto, Bytes, GiB, MiB };
 
+// Bare-number "auto" heuristic threshold: bare values below it are GiB,
+// larger are bytes. The SINGLE source of truth is Python
+// utils/quantity.py:BARE_AUTO_GIB_THRESHOLD — package import pushes it here
+// (bindings set_bare_auto_gib_threshold), so the C++ fast path and the
+// Python fallback can never drift per request path.
+inline std::atomic<int64_t>& bare_auto_gib_threshold() {
+  static std::atomic<int64_t> v{8192};
+  return v;
+}
+
 inline int64_t memory_bytes(const egsjson::Value& v, BareUnit bare) {
   int64_t raw = 0;
   bool had_suffix = false;
@@ -105,9 +115,10 @@ inline int64_t memory_bytes(const egsjson::Value& v, BareUnit bare) {
     case BareUnit::Bytes: return raw;
     case BareUnit::GiB: return raw * (1LL << 30);
     case BareUnit::MiB: return raw * (1LL << 20);
-    case BareUnit::Auto:
-      // utils/quantity.py: bare values < 8192 are GiB, larger are bytes.
-      return (raw > 0 && raw < 8192) ? raw * (1LL << 30) : raw;
+    case BareUnit::Auto: {
+      int64_t thr = bare_auto_gib_threshold().load(std::memory_order_relaxed);
+      return (raw > 0 && raw < thr) ? raw * (1LL << 30) : raw;
+    }
   }
   return raw;
 }

@@ -200,18 +200,28 @@ def node_devices(node: Node, bare_unit: str = "auto") -> List[Any]:
     if inv_raw:
         try:
             inv = json.loads(inv_raw)
+            # Device list position == PHYSICAL card index. The vector is
+            # keyed by each card's published "index", and any gap (a card an
+            # agent omitted, e.g. a sick one gated by health) is filled with
+            # a zero-capacity placeholder — never schedulable, but keeping
+            # every later card at its true physical index. A sick card
+            # published as a zero-capacity entry (agent.publish_with_health)
+            # lands here identically.
+            by_index: Dict[int, Dict[str, Any]] = {}
+            for card in inv.get("cards", []) or []:
+                by_index[int(card.get("index", len(by_index)))] = card
             devices = []
-            # device list position == physical card index: sort defensively
-            # (an agent publishing cards out of order must not skew the
-            # annotation index -> physical card mapping)
-            cards = sorted(inv.get("cards", []),
-                           key=lambda c: int(c.get("index", 0)))
-            for card in cards:
-                mem = int(card.get("memory_bytes", t.MI355X_MEMORY_BYTES))
-                devices.append(core.Device(
-                    core_total=int(card.get("core", t.GPU_CORE_EACH_CARD)),
-                    core_avail=int(card.get("core", t.GPU_CORE_EACH_CARD)),
-                    mem_total=mem, mem_avail=mem))
+            if by_index:
+                for i in range(max(by_index) + 1):
+                    card = by_index.get(i)
+                    if card is None or card.get("sick"):
+                        devices.append(core.Device(core_total=0, core_avail=0,
+                                                   mem_total=0, mem_avail=0))
+                        continue
+                    mem = int(card.get("memory_bytes", t.MI355X_MEMORY_BYTES))
+                    cr = int(card.get("core", t.GPU_CORE_EACH_CARD))
+                    devices.append(core.Device(core_total=cr, core_avail=cr,
+                                               mem_total=mem, mem_avail=mem))
             if devices:
                 return devices
         except (ValueError, TypeError):

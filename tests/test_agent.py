@@ -9,6 +9,7 @@ from elastic_gpu_scheduler_amd.agent import topology as topo
 from elastic_gpu_scheduler_amd.agent.agent import NodeAgent
 from elastic_gpu_scheduler_amd.k8s import objects as obj
 from elastic_gpu_scheduler_amd.k8s.client import FakeKubeClient
+from elastic_gpu_scheduler_amd._native import core
 
 GiB = 1024**3
 
@@ -172,8 +173,10 @@ def test_parse_amd_smi_topology():
 
 
 def test_publish_with_health_excludes_sick_cards(monkeypatch):
-    """A card failing the HBM health gate disappears from the published
-    inventory and allocatable, so the scheduler stops placing onto it."""
+    """A card failing the HBM health gate is published as a ZERO-CAPACITY
+    placeholder: the scheduler stops placing onto it, but every other card
+    keeps its physical index (list position == physical card index is the
+    contract the device-index annotations rely on)."""
     cards = [{"index": i, "memory_bytes": 288 * GiB, "core": 100}
              for i in range(4)]
     monkeypatch.setattr(inv, "discover", lambda prefer="auto": cards)
@@ -192,6 +195,14 @@ def test_publish_with_health_excludes_sick_cards(monkeypatch):
     assert out["sick"] == [1]
     node = client.get_node("gpu-node")
     devs = obj.node_devices(node)
-    assert len(devs) == 3
+    # All four physical slots present; the sick one is unschedulable.
+    assert len(devs) == 4
+    assert not devs[1].schedulable() and not devs[1].whole_free()
+    assert [d.schedulable() for d in devs] == [True, False, True, True]
+    # Allocatable counts only healthy capacity.
     assert node["status"]["allocatable"]["elasticgpu.io/gpu-core"] == "300"
     assert node["status"]["allocatable"]["amd.com/gpu"] == "3"
+    # A whole-card pod never lands on the sick card: its index is skipped.
+    feasible, option, _ = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=3)], "binpack")
+    assert feasible and sorted(option.allocated[0]) == [0, 2, 3]

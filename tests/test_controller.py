@@ -104,8 +104,41 @@ def test_resync_evicts_vanished_pods():
         # vanish without a DELETE event: remove directly from the store
         with client._mu:
             client._pods.clear()
+        # Eviction needs TWO consecutive relist misses (a single-relist
+        # evict races pods bound between list_pods() and the sweep).
+        ctrl.resync_once()
+        assert any(d.core_avail == 60 for d in sch.state.node_devices("n1"))
         ctrl.resync_once()
         assert all(d.core_avail == 100 for d in sch.state.node_devices("n1"))
+    finally:
+        ctrl.stop()
+
+
+def test_resync_snapshot_race_does_not_evict_fresh_pod():
+    """a pod bound between list_pods() (the snapshot) and the
+    sweep must NOT be forgotten — one stale relist marks it missing, and a
+    later relist that sees it clears the mark."""
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=40, memory=GiB))
+        sch.assume(["n1"], pod)
+        sch.bind("n1", pod)
+        assert ctrl.wait_idle()
+        # Simulate a list snapshot taken BEFORE this pod was committed.
+        real_list = client.list_pods
+        client.list_pods = lambda *a, **k: []
+        ctrl.resync_once()
+        # Still accounted after one stale relist.
+        assert any(d.core_avail == 60 for d in sch.state.node_devices("n1"))
+        # The next relist sees the pod again: the missing mark is cleared...
+        client.list_pods = real_list
+        ctrl.resync_once()
+        assert any(d.core_avail == 60 for d in sch.state.node_devices("n1"))
+        # ...so even another stale relist does not evict it.
+        client.list_pods = lambda *a, **k: []
+        ctrl.resync_once()
+        assert any(d.core_avail == 60 for d in sch.state.node_devices("n1"))
     finally:
         ctrl.stop()
 

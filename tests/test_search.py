@@ -154,3 +154,58 @@ def test_huge_subset_space_stays_bounded():
         devs, [core.GPUUnit(10, 0, 0)], "binpack", 0, [])
     assert feasible and len(set(opt.allocated[0])) == 10
     assert leaves <= 4096  # C(20,10)=184,756 raw combinations
+
+
+def test_doomed_last_container_fails_fast():
+    """a multi-container pod whose LAST container can never fit
+    (even on an empty node) must not explore every placement of the earlier
+    containers while holding the node mutex — the pre-check rejects it with
+    zero leaf evaluations."""
+    import time
+
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(64)]
+    # Non-uniform topology disables the symmetric-branch dedupe.
+    hops = [[0 if i == j else 1 + ((i + j) % 3) for j in range(64)]
+            for i in range(64)]
+    units = [core.GPUUnit(0, 10, GiB) for _ in range(6)]
+    units.append(core.GPUUnit(0, 10, 10_000 * GiB))  # never fits any card
+    t0 = time.monotonic()
+    feasible, _, leaves = core.search_placement(devs, units, "binpack", 0, hops)
+    dt = time.monotonic() - t0
+    assert not feasible
+    assert leaves == 0
+    assert dt < 0.1  # fail-fast, not a cards^containers walk
+
+
+def test_oversized_whole_card_ask_fails_fast():
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB) for _ in range(8)]
+    feasible, _, leaves = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=9)], "binpack")
+    assert not feasible and leaves == 0
+
+
+def test_zero_capacity_placeholder_never_scheduled():
+    """Zero-capacity devices (sick-card placeholders) are skipped by both
+    whole-card and fractional placement."""
+    devs = [core.Device(100, 100, 288 * GiB, 288 * GiB),
+            core.Device(0, 0, 0, 0),
+            core.Device(100, 100, 288 * GiB, 288 * GiB)]
+    feasible, opt, _ = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=2)], "binpack")
+    assert feasible and sorted(opt.allocated[0]) == [0, 2]
+    feasible, opt, _ = core.search_placement(
+        devs, [core.GPUUnit(0, 30, GiB), core.GPUUnit(0, 30, GiB)], "spread")
+    assert feasible
+    assert all(c != 1 for a in opt.allocated for c in a)
+    # And a request needing more cards than the healthy count is infeasible.
+    feasible, _, _ = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=3)], "binpack")
+    assert not feasible
+
+
+def test_bare_auto_threshold_single_source():
+    """The C++ fast path's bare-number GiB/bytes threshold is pushed from
+    utils/quantity.py at import — one definition, no per-path drift."""
+    from elastic_gpu_scheduler_amd.utils import quantity
+
+    assert core.get_bare_auto_gib_threshold() == quantity.BARE_AUTO_GIB_THRESHOLD
