@@ -152,7 +152,14 @@ class Controller:
                 log.exception("resync failed")
 
     def resync_once(self) -> None:
-        pods = self.client.list_pods()
+        # Relist only ASSUMED pods (label selector, server-side): every pod
+        # the scheduler accounts carries elasticgpu.io/assumed=true (set at
+        # bind), so this is sufficient for both replay and eviction — and at
+        # real cluster scale it avoids relisting every pod in the cluster
+        # each period (the reference pays that via its informer's full
+        # resync, controller.go:24).
+        from elastic_gpu_scheduler_amd.utils import types as t
+        pods = self.client.list_pods(label_selector={t.EGPU_ASSUMED: "true"})
         live = set()
         for pod in pods:
             if not obj.is_gpu_pod(pod):

@@ -178,16 +178,22 @@ def test_watch_pods_streams_and_reconnects():
             connections["n"] += 1
             if connections["n"] == 1:
                 body = (json.dumps({"type": "ADDED",
-                                    "object": {"metadata": {"name": "w1"}}})
+                                    "object": {"metadata": {"name": "w1",
+                                               "resourceVersion": "11"}}})
                         + "\n" +
                         json.dumps({"type": "MODIFIED",
-                                    "object": {"metadata": {"name": "w1"}}})
+                                    "object": {"metadata": {"name": "w1",
+                                               "resourceVersion": "12"}}})
                         + "\n")
             else:
                 body = json.dumps({"type": "DELETED",
-                                   "object": {"metadata": {"name": "w1"}}}) + "\n"
+                                   "object": {"metadata": {"name": "w1",
+                                              "resourceVersion": "13"}}}) + "\n"
             return httpx.Response(200, content=body.encode())
-        return httpx.Response(404)
+        # initial relist establishing the snapshot + RV to watch from
+        return httpx.Response(200, json={
+            "kind": "PodList", "metadata": {"resourceVersion": "10"},
+            "items": []})
 
     client = RealKubeClient("https://apiserver", token="t",
                             transport=httpx.MockTransport(handler))
@@ -208,3 +214,226 @@ def test_watch_pods_streams_and_reconnects():
     assert ("MODIFIED", "w1") in events_delivered
     assert ("DELETED", "w1") in events_delivered  # arrived via reconnect
     assert connections["n"] >= 2
+
+
+def test_watch_resumes_from_resource_version():
+    """VERDICT r1 #3: a dropped watch RESUMES from the last seen
+    resourceVersion — the event created during the gap is delivered by the
+    resumed watch, and no second full relist happens."""
+    import threading
+
+    lists = {"n": 0}
+    watch_rvs = []
+    delivered = []
+    done = threading.Event()
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        if request.url.params.get("watch") == "true":
+            watch_rvs.append(request.url.params.get("resourceVersion"))
+            if len(watch_rvs) == 1:
+                # one event, then the stream drops
+                body = json.dumps({"type": "ADDED",
+                                   "object": {"metadata": {"name": "a",
+                                              "resourceVersion": "21"}}}) + "\n"
+            else:
+                # resumed watch delivers what happened during the gap
+                body = json.dumps({"type": "ADDED",
+                                   "object": {"metadata": {"name": "gap-pod",
+                                              "resourceVersion": "22"}}}) + "\n"
+            return httpx.Response(200, content=body.encode())
+        lists["n"] += 1
+        return httpx.Response(200, json={
+            "kind": "PodList", "metadata": {"resourceVersion": "20"},
+            "items": [{"metadata": {"name": "seed", "resourceVersion": "19"}}]})
+
+    client = RealKubeClient("https://apiserver", token="t",
+                            transport=httpx.MockTransport(handler))
+
+    def on_event(etype, obj):
+        name = obj.get("metadata", {}).get("name")
+        delivered.append((etype, name))
+        if name == "gap-pod":
+            done.set()
+
+    unsubscribe = client.watch_pods(on_event)
+    try:
+        assert done.wait(timeout=10), delivered
+    finally:
+        unsubscribe()
+    # initial snapshot delivered as synthetic MODIFIED
+    assert ("MODIFIED", "seed") in delivered
+    assert ("ADDED", "gap-pod") in delivered  # nothing lost in the gap
+    assert lists["n"] == 1, "reconnect must NOT trigger a relist storm"
+    assert watch_rvs[0] == "20"   # watch starts from the list's RV
+    assert watch_rvs[1] == "21"   # resume from the last DELIVERED event's RV
+
+
+def test_watch_handles_410_gone_with_single_relist():
+    """An ERROR event with Status code 410 (RV fell out of etcd's window)
+    triggers exactly one relist, then watching resumes from the fresh RV."""
+    import threading
+
+    lists = {"n": 0}
+    watch_calls = {"n": 0}
+    delivered = []
+    done = threading.Event()
+
+    def handler(request: httpx.Request) -> httpx.Response:
+        if request.url.params.get("watch") == "true":
+            watch_calls["n"] += 1
+            if watch_calls["n"] == 1:
+                body = json.dumps({
+                    "type": "ERROR",
+                    "object": {"kind": "Status", "code": 410,
+                               "reason": "Expired"}}) + "\n"
+            else:
+                body = json.dumps({"type": "ADDED",
+                                   "object": {"metadata": {"name": "fresh",
+                                              "resourceVersion": "31"}}}) + "\n"
+            return httpx.Response(200, content=body.encode())
+        lists["n"] += 1
+        return httpx.Response(200, json={
+            "kind": "PodList",
+            "metadata": {"resourceVersion": str(30 + lists["n"])},
+            "items": []})
+
+    client = RealKubeClient("https://apiserver", token="t",
+                            transport=httpx.MockTransport(handler))
+
+    def on_event(etype, obj):
+        delivered.append((etype, obj.get("metadata", {}).get("name")))
+        if obj.get("metadata", {}).get("name") == "fresh":
+            done.set()
+
+    unsubscribe = client.watch_pods(on_event)
+    try:
+        assert done.wait(timeout=10), delivered
+    finally:
+        unsubscribe()
+    assert lists["n"] == 2  # initial + exactly one post-410 relist
+    assert not any(n == "Status" for _, n in delivered)  # ERROR not delivered
+
+
+def test_from_kubeconfig_client_certificates():
+    """A kind/minikube-style kubeconfig (inline base64 client cert/key +
+    CA data) must configure mTLS — r1 read only user.token and could not
+    connect to a default kind cluster at all (VERDICT r1 #2)."""
+    import base64
+
+    fake_ca = b"-----BEGIN CERTIFICATE-----\nZmFrZQ==\n-----END CERTIFICATE-----\n"
+    fake_crt = b"-----BEGIN CERTIFICATE-----\nY2VydA==\n-----END CERTIFICATE-----\n"
+    fake_key = b"-----BEGIN RSA PRIVATE KEY-----\na2V5\n-----END RSA PRIVATE KEY-----\n"
+    cfg = {
+        "current-context": "kind-kind",
+        "contexts": [{"name": "kind-kind",
+                      "context": {"cluster": "kind", "user": "kind-user"}}],
+        "clusters": [{"name": "kind", "cluster": {
+            "server": "https://127.0.0.1:6443",
+            "certificate-authority-data":
+                base64.b64encode(fake_ca).decode()}}],
+        "users": [{"name": "kind-user", "user": {
+            "client-certificate-data": base64.b64encode(fake_crt).decode(),
+            "client-key-data": base64.b64encode(fake_key).decode()}}],
+    }
+    captured = {}
+
+    import elastic_gpu_scheduler_amd.k8s.client as client_mod
+
+    class CapturingClient:
+        def __init__(self, **kw):
+            captured.update(kw)
+
+        def close(self):
+            pass
+
+    real_httpx_client = httpx.Client
+    httpx.Client = lambda **kw: CapturingClient(**kw)
+    try:
+        c = client_mod.RealKubeClient.from_kubeconfig(cfg)
+    finally:
+        httpx.Client = real_httpx_client
+    try:
+        assert captured["base_url"] == "https://127.0.0.1:6443"
+        cert = captured["cert"]
+        assert cert is not None
+        with open(cert[0], "rb") as f:
+            assert f.read() == fake_crt
+        with open(cert[1], "rb") as f:
+            assert f.read() == fake_key
+        with open(captured["verify"], "rb") as f:
+            assert f.read() == fake_ca
+        assert "Authorization" not in captured["headers"]
+    finally:
+        c.close()
+    # temp material is wiped on close
+    import os
+    assert not os.path.exists(cert[0])
+
+
+def test_from_kubeconfig_token_and_ca_paths(tmp_path):
+    ca = tmp_path / "ca.crt"
+    ca.write_text("ca")
+    tok = tmp_path / "token"
+    tok.write_text("sekrit\n")
+    cfg = {
+        "current-context": "c",
+        "contexts": [{"name": "c", "context": {"cluster": "cl", "user": "u"}}],
+        "clusters": [{"name": "cl", "cluster": {
+            "server": "https://h:6443", "certificate-authority": "ca.crt"}}],
+        "users": [{"name": "u", "user": {"tokenFile": str(tok)}}],
+    }
+    captured = {}
+
+    class CapturingClient:
+        def __init__(self, **kw):
+            captured.update(kw)
+
+        def close(self):
+            pass
+
+    real_httpx_client = httpx.Client
+    httpx.Client = lambda **kw: CapturingClient(**kw)
+    try:
+        c = RealKubeClient.from_kubeconfig(cfg, base_dir=tmp_path)
+    finally:
+        httpx.Client = real_httpx_client
+    assert captured["headers"]["Authorization"] == "Bearer sekrit"
+    assert captured["verify"] == str(ca)
+    c.close()
+
+
+def test_from_kubeconfig_exec_plugin(tmp_path):
+    """ExecCredential plugin auth: the plugin's stdout token is used."""
+    plugin = tmp_path / "plugin.sh"
+    plugin.write_text(
+        "#!/bin/sh\n"
+        "echo '{\"apiVersion\":\"client.authentication.k8s.io/v1\","
+        "\"kind\":\"ExecCredential\",\"status\":{\"token\":\"exec-token\"}}'\n")
+    plugin.chmod(0o755)
+    cfg = {
+        "current-context": "c",
+        "contexts": [{"name": "c", "context": {"cluster": "cl", "user": "u"}}],
+        "clusters": [{"name": "cl", "cluster": {
+            "server": "https://h:6443", "insecure-skip-tls-verify": True}}],
+        "users": [{"name": "u", "user": {"exec": {
+            "apiVersion": "client.authentication.k8s.io/v1",
+            "command": str(plugin)}}}],
+    }
+    captured = {}
+
+    class CapturingClient:
+        def __init__(self, **kw):
+            captured.update(kw)
+
+        def close(self):
+            pass
+
+    real_httpx_client = httpx.Client
+    httpx.Client = lambda **kw: CapturingClient(**kw)
+    try:
+        c = RealKubeClient.from_kubeconfig(cfg)
+    finally:
+        httpx.Client = real_httpx_client
+    assert captured["headers"]["Authorization"] == "Bearer exec-token"
+    assert captured["verify"] is False
+    c.close()
