@@ -320,9 +320,27 @@ class RealKubeClient(KubeClient):
         if token:
             headers["Authorization"] = f"Bearer {token}"
         self._tmpdir = _tmpdir  # holds decoded *-data material alive
+        # httpx 0.28 silently DROPS cert=(crt, key) when verify is a CA path
+        # (deprecated combination) — client certificates then never reach the
+        # TLS handshake and every request is anonymous. Build the SSLContext
+        # ourselves so mTLS actually happens (caught by the strict-apiserver
+        # e2e suite; a permissive mock never noticed).
+        if cert is not None:
+            import ssl
+
+            if verify is False:
+                ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_CLIENT)
+                ctx.check_hostname = False
+                ctx.verify_mode = ssl.CERT_NONE
+            else:
+                ctx = ssl.create_default_context(
+                    cafile=verify if isinstance(verify, str) else None)
+            if cert is not None:
+                ctx.load_cert_chain(cert[0], cert[1])
+            verify = ctx
         self._client = httpx.Client(base_url=base_url, headers=headers,
                                     verify=verify, timeout=30.0,
-                                    transport=transport, cert=cert)
+                                    transport=transport)
         self._watch_stop = threading.Event()
 
     def close(self) -> None:

@@ -314,15 +314,19 @@ def test_watch_handles_410_gone_with_single_relist():
     assert not any(n == "Status" for _, n in delivered)  # ERROR not delivered
 
 
-def test_from_kubeconfig_client_certificates():
+def test_from_kubeconfig_client_certificates(tmp_path):
     """A kind/minikube-style kubeconfig (inline base64 client cert/key +
     CA data) must configure mTLS — r1 read only user.token and could not
-    connect to a default kind cluster at all (VERDICT r1 #2)."""
+    connect to a default kind cluster at all. Real PKI
+    material: the client loads it into an SSLContext at construction."""
     import base64
 
-    fake_ca = b"-----BEGIN CERTIFICATE-----\nZmFrZQ==\n-----END CERTIFICATE-----\n"
-    fake_crt = b"-----BEGIN CERTIFICATE-----\nY2VydA==\n-----END CERTIFICATE-----\n"
-    fake_key = b"-----BEGIN RSA PRIVATE KEY-----\na2V5\n-----END RSA PRIVATE KEY-----\n"
+    from elastic_gpu_scheduler_amd.testing import generate_pki
+
+    pki = generate_pki(tmp_path)
+    fake_ca = open(pki["ca_crt"], "rb").read()
+    fake_crt = open(pki["client_crt"], "rb").read()
+    fake_key = open(pki["client_key"], "rb").read()
     cfg = {
         "current-context": "kind-kind",
         "contexts": [{"name": "kind-kind",
@@ -352,22 +356,26 @@ def test_from_kubeconfig_client_certificates():
         c = client_mod.RealKubeClient.from_kubeconfig(cfg)
     finally:
         httpx.Client = real_httpx_client
+    import os
+    import ssl
+
     try:
         assert captured["base_url"] == "https://127.0.0.1:6443"
-        cert = captured["cert"]
-        assert cert is not None
-        with open(cert[0], "rb") as f:
+        # httpx 0.28 drops cert=(crt, key) when verify is a CA path, so the
+        # client must hand httpx a ready SSLContext with the cert loaded.
+        assert isinstance(captured["verify"], ssl.SSLContext)
+        tmp = c._tmpdir.name
+        with open(os.path.join(tmp, "client.crt"), "rb") as f:
             assert f.read() == fake_crt
-        with open(cert[1], "rb") as f:
+        with open(os.path.join(tmp, "client.key"), "rb") as f:
             assert f.read() == fake_key
-        with open(captured["verify"], "rb") as f:
+        with open(os.path.join(tmp, "ca.crt"), "rb") as f:
             assert f.read() == fake_ca
         assert "Authorization" not in captured["headers"]
     finally:
         c.close()
     # temp material is wiped on close
-    import os
-    assert not os.path.exists(cert[0])
+    assert not os.path.exists(tmp)
 
 
 def test_from_kubeconfig_token_and_ca_paths(tmp_path):
