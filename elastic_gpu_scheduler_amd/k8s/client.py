@@ -504,6 +504,15 @@ class RealKubeClient(KubeClient):
             f"/api/v1/namespaces/{meta.get('namespace', 'default')}/pods/"
             f"{meta['name']}", content=json.dumps(pod)))
 
+    def create_pod(self, pod: Pod) -> Pod:
+        ns = pod.get("metadata", {}).get("namespace", "default")
+        return self._check(self._client.post(
+            f"/api/v1/namespaces/{ns}/pods", content=json.dumps(pod)))
+
+    def delete_pod(self, namespace: str, name: str) -> None:
+        self._check(self._client.delete(
+            f"/api/v1/namespaces/{namespace}/pods/{name}"))
+
     def bind_pod(self, namespace: str, name: str, node: str) -> None:
         binding = {
             "apiVersion": "v1",
@@ -517,6 +526,10 @@ class RealKubeClient(KubeClient):
 
     def get_node(self, name: str) -> Node:
         return self._check(self._client.get(f"/api/v1/nodes/{name}"))
+
+    def create_node(self, node: Node) -> Node:
+        return self._check(self._client.post("/api/v1/nodes",
+                                             content=json.dumps(node)))
 
     def list_nodes(self) -> List[Node]:
         out = self._check(self._client.get("/api/v1/nodes"))

@@ -147,6 +147,8 @@ class StrictAPIServer:
 
         class Handler(BaseHTTPRequestHandler):
             protocol_version = "HTTP/1.1"
+            disable_nagle_algorithm = True  # else ~40 ms delayed-ACK stalls
+            wbufsize = 65536
 
             # silence default stderr logging
             def log_message(self, fmt, *args):  # noqa: D401
@@ -414,6 +416,20 @@ class StrictAPIServer:
             h._send_json(200, {"kind": "NodeList", "apiVersion": "v1",
                                "metadata": {"resourceVersion": rv},
                                "items": items})
+            return
+        if method == "POST" and path == "/api/v1/nodes":
+            self._require_json_ct(h)
+            node = self._parse_json(h._read_body())
+            name = node.get("metadata", {}).get("name")
+            if not name:
+                raise _ApiError(422, "Invalid", "metadata.name is required")
+            with self._mu:
+                if name in self._nodes:
+                    raise _ApiError(409, "AlreadyExists",
+                                    f'nodes "{name}" already exists')
+                self._stamp_new_locked(node, None)
+                self._nodes[name] = node
+                h._send_json(201, json.loads(json.dumps(node)))
             return
 
         m = self._POD_BIND.match(path)
@@ -697,41 +713,96 @@ class StrictAPIServer:
 
         with self._watch_cv:
             gen = self._watch_gen
+            too_old = False
             if since is not None and since != 0:
                 oldest = self._log[0][0] if self._log else self._rv + 1
-                if since < oldest - 1 and since < self._rv:
-                    # compacted away: Status 410, the "too old" signal
-                    try:
-                        chunk({"type": "ERROR",
-                               "object": _status(
-                                   410, "Expired",
-                                   f"too old resource version: {since} "
-                                   f"({oldest})") | {"status": "Failure"}})
-                    except (BrokenPipeError, ConnectionResetError, OSError):
-                        pass
-                    end()
-                    return
+                too_old = since < oldest - 1 and since < self._rv
             cursor = since if since not in (None, 0) else self._rv
-            last_bookmark = time.monotonic()
+        if too_old:
+            # compacted away: Status 410, the "too old" signal
             try:
-                while gen == self._watch_gen:
-                    sent = False
-                    for rv, etype, snapshot in self._log:
-                        if rv > cursor:
-                            chunk({"type": etype, "object": snapshot})
-                            cursor = rv
-                            sent = True
-                    if not sent:
-                        self._watch_cv.wait(timeout=self.bookmark_interval)
-                    if bookmarks and \
-                            time.monotonic() - last_bookmark >= \
-                            self.bookmark_interval:
-                        chunk({"type": "BOOKMARK",
-                               "object": {"kind": "Pod", "apiVersion": "v1",
-                                          "metadata": {"resourceVersion":
-                                                       str(self._rv)}}})
-                        cursor = max(cursor, self._rv)
-                        last_bookmark = time.monotonic()
+                chunk({"type": "ERROR",
+                       "object": _status(
+                           410, "Expired",
+                           f"too old resource version: {since}")
+                       | {"status": "Failure"}})
             except (BrokenPipeError, ConnectionResetError, OSError):
-                return  # client went away
-            end()  # dropped by drop_watches(): clean end-of-stream
+                pass
+            end()
+            return
+        last_bookmark = time.monotonic()
+        try:
+            while True:
+                # Collect pending events UNDER the lock; write them to the
+                # (possibly slow) client socket OUTSIDE it — a stalled watch
+                # reader must never block apiserver writes.
+                with self._watch_cv:
+                    if gen != self._watch_gen:
+                        break
+                    pending = [(rv, etype, snapshot)
+                               for rv, etype, snapshot in self._log
+                               if rv > cursor]
+                    if not pending:
+                        self._watch_cv.wait(timeout=self.bookmark_interval)
+                        pending = [(rv, etype, snapshot)
+                                   for rv, etype, snapshot in self._log
+                                   if rv > cursor]
+                        if gen != self._watch_gen:
+                            break
+                    rv_now = self._rv
+                for rv, etype, snapshot in pending:
+                    chunk({"type": etype, "object": snapshot})
+                    cursor = rv
+                if bookmarks and \
+                        time.monotonic() - last_bookmark >= \
+                        self.bookmark_interval:
+                    chunk({"type": "BOOKMARK",
+                           "object": {"kind": "Pod", "apiVersion": "v1",
+                                      "metadata": {"resourceVersion":
+                                                   str(rv_now)}}})
+                    cursor = max(cursor, rv_now)
+                    last_bookmark = time.monotonic()
+        except (BrokenPipeError, ConnectionResetError, OSError):
+            return  # client went away
+        end()  # dropped by drop_watches(): clean end-of-stream
+
+
+def main(argv=None) -> int:
+    """Run the strict apiserver as its OWN process (bench/e2e use): a real
+    apiserver is out-of-process, and keeping it in-process would serialize
+    it behind the scheduler's GIL, measuring the mock instead of the
+    scheduler. Prints one line `READY <port>` on stdout when serving."""
+    import argparse
+    import sys
+
+    p = argparse.ArgumentParser(description=main.__doc__)
+    p.add_argument("--pki", required=True,
+                   help="directory with/for CA+server+client certs "
+                        "(generated if missing)")
+    p.add_argument("--port", type=int, default=0)
+    p.add_argument("--token", default=None)
+    args = p.parse_args(argv)
+
+    d = Path(args.pki)
+    if not (d / "server.crt").exists():
+        pki = generate_pki(d)
+    else:
+        pki = {"ca_crt": str(d / "ca.crt"), "ca_key": str(d / "ca.key"),
+               "server_crt": str(d / "server.crt"),
+               "server_key": str(d / "server.key"),
+               "client_crt": str(d / "client.crt"),
+               "client_key": str(d / "client.key")}
+    server = StrictAPIServer(pki, token=args.token, port=args.port).start()
+    print(f"READY {server.port}", flush=True)
+    try:
+        while True:
+            time.sleep(3600)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        server.stop()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

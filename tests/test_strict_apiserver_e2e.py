@@ -13,6 +13,7 @@ import base64
 import json
 import threading
 import time
+from pathlib import Path
 
 import httpx
 import pytest
@@ -363,3 +364,21 @@ def _count_pod_lists(apiserver, monkeypatch):
         return orig(h, params)
 
     monkeypatch.setattr(apiserver, "_serve_pod_list", counting)
+
+
+def test_real_wire_bench_smoke():
+    """benchmarks/e2e_real_wire.py stays runnable (tiny shape): the measured
+    config-1 number in profiles/r02_results.md comes from this harness."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "benchmarks/e2e_real_wire.py", "--steps", "1",
+         "--warmup", "0", "--batch", "8", "--concurrency", "1"],
+        capture_output=True, text=True, timeout=240,
+        cwd=str(Path(__file__).resolve().parent.parent))
+    assert out.returncode == 0, out.stderr[-2000:]
+    result = json.loads(out.stdout.strip().splitlines()[-1])
+    assert result["metric"] == "pods_scheduled_per_sec_real_wire"
+    assert result["value"] > 0
+    assert result["config"]["bind_retries"] == 0
