@@ -62,7 +62,7 @@ def build_core(force: bool = False) -> Path:
             + _py_includes()
             + ["-I" + str(CORE_SRC)]
             + [str(s) for s in sorted(CORE_SRC.glob("*.cc"))]
-            + ["-o", str(out)]
+            + ["-o", str(out), "-lssl", "-lcrypto"]  # native-server TLS
         )
         _run(cmd)
     return out

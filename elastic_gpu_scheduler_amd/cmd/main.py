@@ -57,9 +57,13 @@ def build_parser() -> argparse.ArgumentParser:
                         "for the lease instead of double-booking GPUs (the "
                         "reference supports only replicas=1)")
     p.add_argument("--tls-cert", default="", metavar="PEM",
-                   help="serve HTTPS (extender enableHTTPS) — uvicorn "
-                        "front end only")
+                   help="serve HTTPS (extender enableHTTPS); works on BOTH "
+                        "front ends — the native C++ server terminates TLS "
+                        "with OpenSSL and keeps the GIL-free fast path")
     p.add_argument("--tls-key", default="", metavar="PEM")
+    p.add_argument("--tls-client-ca", default="", metavar="PEM",
+                   help="require+verify client certificates (mTLS), native "
+                        "front end")
     p.add_argument("--leader-identity",
                    default=os.environ.get("POD_NAME", "") or os.uname().nodename)
     p.add_argument("--log-level", default="info")
@@ -151,13 +155,14 @@ def main(argv=None) -> int:
 
     log.info("listening on %s:%d (policy=%s mode=%s server=%s)", args.host,
              args.port, args.priority, args.mode, args.server)
-    if args.tls_cert and args.server == "native":
-        log.warning("TLS requires --server uvicorn; switching front end")
-        args.server = "uvicorn"
     if args.server == "native":
         from elastic_gpu_scheduler_amd.server.native import NativeFrontend
 
-        fe = NativeFrontend(app, host=args.host, port=args.port)
+        # TLS (extender enableHTTPS) is terminated by OpenSSL inside the
+        # C++ server — the GIL-free fast path is kept under HTTPS.
+        fe = NativeFrontend(app, host=args.host, port=args.port,
+                            tls_cert=args.tls_cert, tls_key=args.tls_key,
+                            tls_client_ca=args.tls_client_ca)
         fe.start()
         stop_event.wait()
         fe.stop()

@@ -32,7 +32,10 @@ class NativeExtenderServer {
  public:
   NativeExtenderServer(std::shared_ptr<ClusterState> state,
                        const std::string& bare_unit, const std::string& host,
-                       int port, py::object fallback)
+                       int port, py::object fallback,
+                       const std::string& tls_cert = "",
+                       const std::string& tls_key = "",
+                       const std::string& tls_client_ca = "")
       : core_(std::make_shared<ExtenderCore>(state, parse_bare_unit(bare_unit))),
         fallback_(std::move(fallback)) {
     auto core = core_;
@@ -60,8 +63,16 @@ class NativeExtenderServer {
                      "\"}";
       }
     };
-    server_ = std::make_unique<egshttp::HttpServer>(host, port, handler);
+    egshttp::TlsConfig tls;
+    tls.cert_file = tls_cert;
+    tls.key_file = tls_key;
+    tls.client_ca_file = tls_client_ca;
+    server_ = std::make_unique<egshttp::HttpServer>(host, port, handler,
+                                                    /*max_connections=*/512,
+                                                    std::move(tls));
   }
+
+  bool tls_enabled() const { return server_->tls_enabled(); }
 
   ~NativeExtenderServer() { stop(); }
 
@@ -96,10 +107,14 @@ class NativeExtenderServer {
 void bind_native_server(py::module_& m) {
   py::class_<NativeExtenderServer>(m, "NativeExtenderServer")
       .def(py::init<std::shared_ptr<ClusterState>, const std::string&,
-                    const std::string&, int, py::object>(),
+                    const std::string&, int, py::object, const std::string&,
+                    const std::string&, const std::string&>(),
            py::arg("state"), py::arg("bare_unit") = "auto",
            py::arg("host") = "0.0.0.0", py::arg("port") = 0,
-           py::arg("fallback"))
+           py::arg("fallback"), py::arg("tls_cert") = "",
+           py::arg("tls_key") = "", py::arg("tls_client_ca") = "")
+      .def_property_readonly("tls_enabled",
+                             &NativeExtenderServer::tls_enabled)
       .def("start", &NativeExtenderServer::start,
            py::call_guard<py::gil_scoped_release>())
       .def("stop", &NativeExtenderServer::stop,

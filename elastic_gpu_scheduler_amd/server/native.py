@@ -15,12 +15,19 @@ from elastic_gpu_scheduler_amd.server.app import ExtenderApp
 
 
 class NativeFrontend:
-    def __init__(self, app: ExtenderApp, host: str = "0.0.0.0", port: int = 0):
+    def __init__(self, app: ExtenderApp, host: str = "0.0.0.0", port: int = 0,
+                 tls_cert: str = "", tls_key: str = "",
+                 tls_client_ca: str = ""):
+        """tls_cert/tls_key serve HTTPS (extender enableHTTPS) with OpenSSL
+        terminating TLS inside the C++ server — the GIL-free fast path is
+        kept (r1 silently fell back to uvicorn under TLS).
+        tls_client_ca additionally requires verified client certs (mTLS)."""
         self.app = app
         sch = app.registry.default
         self.server = core.NativeExtenderServer(
             state=sch.state, bare_unit=sch.bare_unit, host=host, port=port,
-            fallback=self._fallback)
+            fallback=self._fallback, tls_cert=tls_cert, tls_key=tls_key,
+            tls_client_ca=tls_client_ca)
         app.native_server = self.server
 
     def _fallback(self, method: str, path: str, body: bytes):
