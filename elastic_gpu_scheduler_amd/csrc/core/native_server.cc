@@ -138,4 +138,8 @@ void bind_native_server(py::module_& m) {
   m.def("get_bare_auto_gib_threshold", []() {
     return bare_auto_gib_threshold().load(std::memory_order_relaxed);
   });
+
+  // De-herded protocol score (exposed for the Python<->C++ parity test).
+  m.def("jittered_int_score", &jittered_int_score, py::arg("score"),
+        py::arg("uid"), py::arg("node"));
 }

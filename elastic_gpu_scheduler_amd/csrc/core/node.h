@@ -46,15 +46,25 @@ class NodeAllocator {
 
   // Feasibility + placement, cached per pod UID. Returns true if a placement
   // exists (and remembers it for the later Score/Allocate of the same pod).
+  //
+  // Large-cluster fast path (256 nodes x 128 pods ran at
+  // 411 pods/s because every pod re-searched every node): identical request
+  // SHAPES share one cached search result per node, invalidated by a
+  // generation counter that bumps on every device-state change. Unlike the
+  // reference's shape-hash assume cache (allocate.go:30-33 — its only
+  // placement cache, which double-books because Allocate consumes it), this
+  // is a pure memo: the committed path still re-validates under the node
+  // lock (allocate -> fits_locked -> re-search on failure), so two
+  // same-shaped pods can never commit one slot twice.
   bool assume(const std::string& uid, const GPURequest& req, const Rater& rater,
               bool distinct = false) {
     std::lock_guard<std::mutex> g(mu_);
     gc_assumed_locked();
     auto it = assumed_.find(uid);
     if (it != assumed_.end()) return true;
-    auto res = run_search_locked(req, rater, distinct);
+    const SearchResult& res = shape_search_locked(req, rater, distinct);
     if (!res.feasible) return false;
-    assumed_[uid] = {std::move(res.option), now()};
+    assumed_[uid] = {res.option, now()};
     return true;
   }
 
@@ -65,7 +75,7 @@ class NodeAllocator {
     std::lock_guard<std::mutex> g(mu_);
     auto it = assumed_.find(uid);
     if (it != assumed_.end()) return it->second.option.score;
-    auto res = run_search_locked(req, rater, distinct);
+    const SearchResult& res = shape_search_locked(req, rater, distinct);
     if (!res.feasible) return kScoreMin;
     assumed_[uid] = {res.option, now()};
     return res.option.score;
@@ -211,6 +221,34 @@ class NodeAllocator {
     return search_placement(devices_, req, rater, ctx, distinct);
   }
 
+  static uint64_t shape_hash(const GPURequest& req, bool distinct) {
+    uint64_t h = detail::fnv1a(14695981039346656037ULL,
+                               distinct ? 0x9e37ULL : 0x79b9ULL);
+    for (const auto& u : req) {
+      h = detail::fnv1a(h, static_cast<uint64_t>(u.gpu_count));
+      h = detail::fnv1a(h, static_cast<uint64_t>(u.core) | (1ULL << 40));
+      h = detail::fnv1a(h, static_cast<uint64_t>(u.memory) ^ (7ULL << 56));
+    }
+    return h;
+  }
+
+  // Memoised search per request shape; valid only while no device state
+  // changed (generation match). The Rater is fixed per scheduler and the
+  // Random rater's salt is per-node (not per-pod), so same shape + same
+  // generation implies the identical search result.
+  const SearchResult& shape_search_locked(const GPURequest& req,
+                                          const Rater& rater, bool distinct) {
+    uint64_t h = shape_hash(req, distinct);
+    auto it = shape_cache_.find(h);
+    if (it != shape_cache_.end() && it->second.gen == gen_)
+      return it->second.result;
+    if (shape_cache_.size() > 4096) shape_cache_.clear();  // bound memory
+    ShapeEntry& e = shape_cache_[h];
+    e.gen = gen_;
+    e.result = run_search_locked(req, rater, distinct);
+    return e.result;
+  }
+
   bool fits_locked(const GPURequest& req, const GPUOption& option) const {
     std::vector<Device> copy = devices_;
     for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
@@ -233,6 +271,7 @@ class NodeAllocator {
   }
 
   void transact_locked(const GPURequest& req, const GPUOption& option) {
+    ++gen_;  // device state changes: shape-cache entries go stale
     for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
       const GPUUnit& u = req[c];
       for (int idx : option.allocated[c]) {
@@ -249,6 +288,7 @@ class NodeAllocator {
   }
 
   void cancel_locked(const GPURequest& req, const GPUOption& option) {
+    ++gen_;
     for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
       const GPUUnit& u = req[c];
       for (int idx : option.allocated[c]) {
@@ -294,7 +334,14 @@ class NodeAllocator {
       assumed_.erase(order[i].second);
   }
 
+  struct ShapeEntry {
+    uint64_t gen = 0;
+    SearchResult result;
+  };
+
   Clock::time_point last_gc_ = Clock::time_point::min();
+  uint64_t gen_ = 1;  // device-state generation (shape-cache validity)
+  std::unordered_map<uint64_t, ShapeEntry> shape_cache_;
   std::string name_;
   std::vector<Device> devices_;
   Topology topo_;

@@ -49,39 +49,117 @@ struct SearchResult {
 
 namespace search_detail {
 
+// Below this many raw k-subsets the enumeration is EXHAUSTIVE (exact
+// best-by-topology ordering); above it, greedy-seeded generation takes
+// over (see below).
+constexpr long long kExhaustiveSubsetCap = 4096;
+
 // Enumerate k-subsets of `free_cards`, best-topology-first, capped.
+//
+// r1 enumerated lexicographically and stopped GENERATING at 1024
+// candidates before sorting, so on a fragmented large node (CPX 64-way)
+// every candidate shared a low-index prefix and the best-locality set
+// could simply never be generated. Now:
+//   * C(n,k) <= kExhaustiveSubsetCap: full enumeration — exact;
+//   * larger: one candidate GREEDY-GROWN FROM EVERY free card (seed ->
+//     repeatedly add the card with the lowest incremental hop cost), then
+//     a single swap-improvement pass per candidate. A seed inside the
+//     best hive always exists, and min-incremental-hop growth stays
+//     inside that hive while it has free cards — so fragmentation cannot
+//     hide the minimum-hop set behind a lexicographic prefix.
+// Everything ties-breaks on card index, so the result is deterministic.
 inline void enumerate_subsets(const std::vector<int>& free_cards, int k,
                               const Topology& topo,
                               std::vector<std::vector<int>>* out) {
   const int n = static_cast<int>(free_cards.size());
   if (k > n) return;
-  std::vector<int> pick;
-  pick.reserve(k);
+  if (k <= 0) return;
   struct Cand {
     std::vector<int> cards;
     int cost;
   };
   std::vector<Cand> cands;
-  // DFS over combinations (lexicographic, deterministic).
-  std::function<void(int)> rec = [&](int start) {
-    if (static_cast<int>(pick.size()) == k) {
-      std::vector<int> cards;
-      cards.reserve(k);
-      for (int i : pick) cards.push_back(free_cards[i]);
-      int cost = topo.set_cost(cards);
-      cands.push_back({std::move(cards), cost});
-      return;
+
+  long long total = 1;  // C(n,k), saturated
+  for (int i = 0; i < k; ++i) {
+    total = total * (n - i) / (i + 1);
+    if (total > kExhaustiveSubsetCap) break;
+  }
+
+  if (total <= kExhaustiveSubsetCap) {
+    std::vector<int> pick;
+    pick.reserve(k);
+    std::function<void(int)> rec = [&](int start) {
+      if (static_cast<int>(pick.size()) == k) {
+        std::vector<int> cards;
+        cards.reserve(k);
+        for (int i : pick) cards.push_back(free_cards[i]);
+        int cost = topo.set_cost(cards);
+        cands.push_back({std::move(cards), cost});
+        return;
+      }
+      for (int i = start; i < n; ++i) {
+        pick.push_back(i);
+        rec(i + 1);
+        pick.pop_back();
+      }
+    };
+    rec(0);
+  } else {
+    for (int s = 0; s < n; ++s) {
+      std::vector<int> set = {free_cards[s]};
+      std::vector<char> used(n, 0);
+      used[s] = 1;
+      while (static_cast<int>(set.size()) < k) {
+        int best = -1;
+        long long best_inc = 0;
+        for (int j = 0; j < n; ++j) {
+          if (used[j]) continue;
+          long long inc = 0;
+          for (int c : set) inc += topo.hops(c, free_cards[j]);
+          if (best < 0 || inc < best_inc) {
+            best = j;
+            best_inc = inc;
+          }
+        }
+        used[best] = 1;
+        set.push_back(free_cards[best]);
+      }
+      std::sort(set.begin(), set.end());
+      // One swap-improvement pass: replace a member with a non-member
+      // when it strictly lowers the pairwise cost (first-improvement,
+      // ascending order -> deterministic).
+      int cost = topo.set_cost(set);
+      for (size_t a = 0; a < set.size(); ++a) {
+        for (int j = 0; j < n; ++j) {
+          int cand = free_cards[j];
+          if (std::find(set.begin(), set.end(), cand) != set.end()) continue;
+          int delta = 0;
+          for (size_t b = 0; b < set.size(); ++b) {
+            if (b == a) continue;
+            delta += topo.hops(set[b], cand) - topo.hops(set[b], set[a]);
+          }
+          if (delta < 0) {
+            set[a] = cand;
+            cost += delta;
+          }
+        }
+      }
+      std::sort(set.begin(), set.end());
+      cands.push_back({std::move(set), cost});
     }
-    if (static_cast<int>(cands.size()) >= kMaxWholeCardCandidates * 8) return;
-    for (int i = start; i < n; ++i) {
-      pick.push_back(i);
-      rec(i + 1);
-      pick.pop_back();
-    }
-  };
-  rec(0);
-  std::stable_sort(cands.begin(), cands.end(),
-                   [](const Cand& a, const Cand& b) { return a.cost < b.cost; });
+    // dedupe identical sets (many seeds converge to the same optimum)
+    std::sort(cands.begin(), cands.end(),
+              [](const Cand& a, const Cand& b) { return a.cards < b.cards; });
+    cands.erase(std::unique(cands.begin(), cands.end(),
+                            [](const Cand& a, const Cand& b) {
+                              return a.cards == b.cards;
+                            }),
+                cands.end());
+  }
+  std::stable_sort(cands.begin(), cands.end(), [](const Cand& a, const Cand& b) {
+    return a.cost != b.cost ? a.cost < b.cost : a.cards < b.cards;
+  });
   int keep = std::min<int>(static_cast<int>(cands.size()), kMaxWholeCardCandidates);
   for (int i = 0; i < keep; ++i) out->push_back(std::move(cands[i].cards));
 }

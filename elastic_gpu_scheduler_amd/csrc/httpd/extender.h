@@ -11,6 +11,7 @@
 #include <algorithm>
 #include <atomic>
 #include <chrono>
+#include <cmath>
 #include <cstdint>
 #include <mutex>
 #include <string>
@@ -204,6 +205,35 @@ inline PodInfo parse_pod(const egsjson::Value& pod, BareUnit bare) {
   return info;
 }
 
+// ---- score -> extender-protocol integer (de-herded) ----
+//
+// The protocol wants ints 0..10, so equally-packed nodes used to tie and
+// every concurrent pod raced for the same node (r1 soak: ~3-4% binpack
+// bind retries, profiles/r01_results.md). Stochastic rounding with a
+// stable per-(pod, node) hash offset splits those ties per pod without
+// changing any ordering that differs by >= 1 score level: floor(s + u)
+// where u in [0,1) is pseudo-random in (uid, node) — deterministic, so
+// Assume/Score/Bind agree, and different pods order tied nodes
+// differently, spreading the herd. MUST stay bit-identical with the
+// Python fallback (server/app.py _jittered_int_score; parity-tested).
+inline int64_t jittered_int_score(double s, const std::string& uid,
+                                  const std::string& node) {
+  uint64_t h = 14695981039346656037ULL;
+  for (unsigned char c : uid) {
+    h ^= c;
+    h *= 1099511628211ULL;
+  }
+  h ^= 0x9e3779b97f4a7c15ULL;
+  for (unsigned char c : node) {
+    h ^= c;
+    h *= 1099511628211ULL;
+  }
+  double u = static_cast<double>(h % 4096) / 4096.0;
+  double v = std::min(std::max(s, 0.0), 10.0);
+  auto out = static_cast<int64_t>(std::floor(v + u));
+  return std::min<int64_t>(out, 10);
+}
+
 // ---- filter -> first-filter timestamp tracker (p50 filter->bind) ----
 
 class FilterTracker {
@@ -333,8 +363,8 @@ class ExtenderCore {
       for (size_t i = 0; i < names.size(); ++i) {
         egsjson::Value e = egsjson::Value::make_object();
         e.set("host", egsjson::Value(names[i]));
-        e.set("score",
-              egsjson::Value(static_cast<int64_t>(std::llround(scores[i]))));
+        e.set("score", egsjson::Value(
+                           jittered_int_score(scores[i], info.uid, names[i])));
         result.push_back(std::move(e));
       }
     }

@@ -112,3 +112,43 @@ def test_filter_and_priorities_parity(idx):
                     "nodenames": ["node-a", "node-b"]}).encode())
     py_scores = {e["host"]: e["score"] for e in json.loads(py_prio)}
     assert nat_scores == py_scores, (containers, nat_scores, py_scores)
+
+
+def test_jittered_int_score_parity():
+    """The de-herded stochastic rounding must be bit-identical between the
+    C++ fast path and the Python fallback — a drift would make priorities
+    depend on which path served the request."""
+    import random
+
+    from elastic_gpu_scheduler_amd._native import core
+    from elastic_gpu_scheduler_amd.server.app import _jittered_int_score
+
+    rng = random.Random(42)
+    for _ in range(500):
+        s = rng.uniform(-1.0, 11.0)
+        uid = f"uid-{rng.randrange(10**9)}"
+        node = f"node-{rng.randrange(1000)}"
+        assert core.jittered_int_score(s, uid, node) == \
+            _jittered_int_score(s, uid, node), (s, uid, node)
+    # protocol range holds at the extremes
+    assert core.jittered_int_score(10.0, "u", "n") == 10
+    assert core.jittered_int_score(0.0, "u", "n") in (0, 1)
+    assert core.jittered_int_score(-5.0, "u", "n") in (0, 1)
+    assert core.jittered_int_score(99.0, "u", "n") == 10
+
+
+def test_jitter_splits_ties_per_pod():
+    """Equally-scored (non-integral) nodes round to DIFFERENT ints for
+    different pods — the herd disperses."""
+    from elastic_gpu_scheduler_amd._native import core
+
+    nodes = [f"node-{i}" for i in range(8)]
+    picks = set()
+    for p in range(32):
+        uid = f"pod-{p}"
+        scores = [core.jittered_int_score(6.5, uid, n) for n in nodes]
+        top = max(scores)
+        picks.add(nodes[scores.index(top)])
+        assert all(s in (6, 7) for s in scores)
+    # 32 pods over 8 tied nodes: far more than one distinct first choice
+    assert len(picks) >= 4, picks

@@ -209,3 +209,90 @@ def test_bare_auto_threshold_single_source():
     from elastic_gpu_scheduler_amd.utils import quantity
 
     assert core.get_bare_auto_gib_threshold() == quantity.BARE_AUTO_GIB_THRESHOLD
+
+
+# ---------------------------------------------------------------------------
+# Fragmentation-proof whole-card subset selection (the
+# r1 lexicographic-prefix candidate cap could hide the minimum-hop set on a
+# fragmented CPX-64 node). Three fragmentation patterns, oracle-checked.
+
+
+def _cpx_hops(n=64, hive=8):
+    """CPX 64-partition topology: 8 hives x 8; intra-hive 1 hop, cross 3."""
+    return [[0 if i == j else (1 if i // hive == j // hive else 3)
+             for j in range(n)] for i in range(n)]
+
+
+def _cpx_devices(free_indexes, n=64):
+    devs = []
+    for i in range(n):
+        if i in free_indexes:
+            devs.append(core.Device(100, 100, 288 * GiB, 288 * GiB))
+        else:
+            devs.append(core.Device(100, 0, 288 * GiB, 0))  # occupied
+    return devs
+
+
+def test_fragmented_cpx_high_index_hive_is_found():
+    """Pattern 1: the ONLY fully-free hive is the LAST one — exactly what
+    the r1 lexicographic prefix cap could never reach. k=8 must land on it."""
+    free = set()
+    for h in range(7):          # hives 0..6: 5 free cards each (fragmented)
+        free.update(range(h * 8, h * 8 + 5))
+    free.update(range(56, 64))  # hive 7: fully free
+    devs = _cpx_devices(free)
+    feasible, opt, _ = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=8)], "binpack", 0, _cpx_hops())
+    assert feasible
+    assert sorted(opt.allocated[0]) == list(range(56, 64))
+
+
+def test_fragmented_cpx_small_set_on_high_hive():
+    """Pattern 2: k=4; the only hive with 4 free cards is hive 7 (indexes
+    56-59). Every other hive offers 3 — a 4-set there costs 3+3*3=... more.
+    C(25,4)=12,650 > exhaustive cap, so this exercises the greedy path."""
+    free = set()
+    for h in range(7):
+        free.update(range(h * 8, h * 8 + 3))  # 3 free per low hive
+    free.update(range(56, 60))                # hive 7: 4 free
+    devs = _cpx_devices(free)
+    feasible, opt, _ = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=4)], "binpack", 0, _cpx_hops())
+    assert feasible
+    assert sorted(opt.allocated[0]) == [56, 57, 58, 59]
+
+
+def test_fragmented_cpx_mixed_set_matches_bruteforce_oracle():
+    """Pattern 3: no hive can satisfy k=5 alone -> the optimum mixes hives.
+    The chosen set's pairwise hop cost must equal the brute-force minimum
+    over all C(20,5)=15,504 subsets."""
+    import itertools
+
+    free = sorted(
+        list(range(16, 20)) +    # hive 2: 4 free
+        list(range(40, 44)) +    # hive 5: 4 free
+        [0, 1, 8, 9, 24, 25, 32, 33, 48, 49, 56, 57])  # 2 per other hive
+    assert len(free) == 20
+    hops = _cpx_hops()
+
+    def cost(cards):
+        return sum(hops[a][b] for a, b in itertools.combinations(cards, 2))
+
+    oracle = min(cost(s) for s in itertools.combinations(free, 5))
+    devs = _cpx_devices(set(free))
+    feasible, opt, _ = core.search_placement(
+        devs, [core.GPUUnit(gpu_count=5)], "binpack", 0, hops)
+    assert feasible
+    chosen = sorted(opt.allocated[0])
+    assert cost(chosen) == oracle, (chosen, cost(chosen), oracle)
+
+
+def test_subset_selection_is_deterministic():
+    free = set(range(0, 64, 2))  # 32 free cards, alternating
+    devs = _cpx_devices(free)
+    results = set()
+    for _ in range(3):
+        _, opt, _ = core.search_placement(
+            devs, [core.GPUUnit(gpu_count=6)], "binpack", 0, _cpx_hops())
+        results.add(tuple(sorted(opt.allocated[0])))
+    assert len(results) == 1
