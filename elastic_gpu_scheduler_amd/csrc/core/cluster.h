@@ -138,6 +138,15 @@ class ClusterState {
     return nodes_.count(name) > 0;
   }
 
+  // One lock acquisition for a whole candidate list (a 256-node filter
+  // would otherwise take the shared lock 256 times per request).
+  bool has_all_nodes(const std::vector<std::string>& names) {
+    std::shared_lock<std::shared_mutex> g(mu_);
+    for (const auto& n : names)
+      if (nodes_.count(n) == 0) return false;
+    return true;
+  }
+
   void remove_node(const std::string& name) {
     std::unique_lock<std::shared_mutex> g(mu_);
     nodes_.erase(name);

@@ -296,8 +296,7 @@ class ExtenderCore {
     std::vector<std::string> names;
     names.reserve(nodenames.as_array().size());
     for (const auto& n : nodenames.as_array()) names.push_back(n.as_string());
-    for (const auto& n : names)
-      if (!state_->has_node(n)) return HandleStatus::NeedFallback;
+    if (!state_->has_all_nodes(names)) return HandleStatus::NeedFallback;
 
     PodInfo info = parse_pod(pod, bare_);
     egsjson::Value out = egsjson::Value::make_object();
@@ -345,8 +344,7 @@ class ExtenderCore {
     std::vector<std::string> names;
     for (const auto& n : args.get("nodenames").as_array())
       names.push_back(n.as_string());
-    for (const auto& n : names)
-      if (!state_->has_node(n)) return HandleStatus::NeedFallback;
+    if (!state_->has_all_nodes(names)) return HandleStatus::NeedFallback;
 
     PodInfo info = parse_pod(pod, bare_);
     egsjson::Array result;

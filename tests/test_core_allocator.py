@@ -188,3 +188,29 @@ class TestDeterminism:
                 c.allocate("n", "warm", frac(50, 10 * GiB))
                 opts.append(c.allocate("n", "p", frac(30, 5 * GiB)).allocated)
             assert opts[0] == opts[1] == opts[2], policy
+
+
+def test_shape_cache_never_double_books():
+    """The per-shape search memo (large-cluster fast path) must stay a pure
+    MEMO: two same-shaped pods may receive the same cached proposal, but
+    commit re-validates — the second bind must fail or land elsewhere, and
+    any state change invalidates the cached verdict (generation bump)."""
+    from elastic_gpu_scheduler_amd._native import core
+
+    state = core.ClusterState("binpack", 0, 2)
+    state.add_node("n", [core.Device(100, 100, GiB, GiB)], [])
+    unit = [core.GPUUnit(0, 60, 0)]  # 60% core: only one fits per card
+    assert state.assume(["n"], "pod-a", unit) == [0]
+    assert state.assume(["n"], "pod-b", unit) == [0]  # same shape, cache hit
+    opt = state.allocate("n", "pod-a", unit)
+    assert opt.allocated[0] == [0]
+    # pod-b's cached proposal no longer fits; allocate must refuse
+    import pytest as _pytest
+    with _pytest.raises(Exception):
+        state.allocate("n", "pod-b", unit)
+    # generation bumped: a fresh same-shaped assume now reports infeasible
+    assert state.assume(["n"], "pod-c", unit) == [1]
+    # release invalidates again: feasible once more
+    state.forget_pod("pod-a")
+    assert state.assume(["n"], "pod-d", unit) == [0]
+    assert state.allocate("n", "pod-d", unit).allocated[0] == [0]
