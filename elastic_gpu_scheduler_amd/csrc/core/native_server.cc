@@ -96,6 +96,28 @@ class NativeExtenderServer {
     return d;
   }
 
+  // Per-verb latency histograms computed IN the GIL-free fast path
+  // (log2 us buckets): verb -> {count, sum_us, buckets: [(le_us, n), ...]}.
+  py::dict latency_histograms() const {
+    py::dict out;
+    auto dump = [](const LatencyHist& h) {
+      py::dict d;
+      d["count"] = h.count.load(std::memory_order_relaxed);
+      d["sum_us"] = h.sum_us.load(std::memory_order_relaxed);
+      py::list buckets;
+      for (int b = 0; b < LatencyHist::kBuckets; ++b) {
+        uint64_t le_us = 1ULL << b;
+        buckets.append(py::make_tuple(
+            le_us, h.buckets[b].load(std::memory_order_relaxed)));
+      }
+      d["buckets"] = std::move(buckets);
+      return d;
+    };
+    out["filter"] = dump(core_->filter_hist);
+    out["priorities"] = dump(core_->priorities_hist);
+    return out;
+  }
+
  private:
   std::shared_ptr<ExtenderCore> core_;
   py::object fallback_;
@@ -122,7 +144,8 @@ void bind_native_server(py::module_& m) {
       .def_property_readonly("port", &NativeExtenderServer::port)
       .def("note_filter", &NativeExtenderServer::note_filter)
       .def("pop_filter_seconds", &NativeExtenderServer::pop_filter_seconds)
-      .def("stats", &NativeExtenderServer::stats);
+      .def("stats", &NativeExtenderServer::stats)
+      .def("latency_histograms", &NativeExtenderServer::latency_histograms);
 
   // JSON codec round-trip (exposed for tests of the native parser).
   m.def("json_roundtrip", [](const std::string& s) {

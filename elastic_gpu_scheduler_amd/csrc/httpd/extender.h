@@ -273,6 +273,46 @@ struct NativeCounters {
   std::atomic<uint64_t> errors{0};
 };
 
+// Lock-free log2 latency histogram for the GIL-free fast path. r1's
+// /debug/profile sampler could only see Python threads, leaving the C++
+// 95% of traffic invisible; these histograms
+// are computed where the work happens and exported through stats() and
+// /metrics. Bucket b counts requests with latency in (2^(b-1), 2^b] us.
+struct LatencyHist {
+  static constexpr int kBuckets = 24;  // 1 us .. ~8.4 s
+  std::atomic<uint64_t> buckets[kBuckets];
+  std::atomic<uint64_t> count{0};
+  std::atomic<uint64_t> sum_us{0};
+
+  LatencyHist() {
+    for (auto& b : buckets) b.store(0, std::memory_order_relaxed);
+  }
+  void record_us(uint64_t us) {
+    int b = us <= 1 ? 0
+                    : std::min<int>(kBuckets - 1,
+                                    64 - __builtin_clzll(us - 1));
+    buckets[b].fetch_add(1, std::memory_order_relaxed);
+    count.fetch_add(1, std::memory_order_relaxed);
+    sum_us.fetch_add(us, std::memory_order_relaxed);
+  }
+};
+
+class ScopedLatency {
+ public:
+  explicit ScopedLatency(LatencyHist* h) : h_(h), t0_(Clock::now()) {}
+  ~ScopedLatency() {
+    auto us = std::chrono::duration_cast<std::chrono::microseconds>(
+                  Clock::now() - t0_)
+                  .count();
+    h_->record_us(static_cast<uint64_t>(us < 0 ? 0 : us));
+  }
+
+ private:
+  using Clock = std::chrono::steady_clock;
+  LatencyHist* h_;
+  Clock::time_point t0_;
+};
+
 enum class HandleStatus { Handled, NeedFallback };
 
 class ExtenderCore {
@@ -282,6 +322,7 @@ class ExtenderCore {
 
   // POST /scheduler/filter. Returns Handled + response JSON, or NeedFallback.
   HandleStatus filter(const std::string& body, std::string* response) {
+    ScopedLatency lat(&filter_hist);
     egsjson::Value args;
     try {
       args = egsjson::parse(body);
@@ -333,6 +374,7 @@ class ExtenderCore {
 
   // POST /scheduler/priorities.
   HandleStatus priorities(const std::string& body, std::string* response) {
+    ScopedLatency lat(&priorities_hist);
     egsjson::Value args;
     try {
       args = egsjson::parse(body);
@@ -373,6 +415,8 @@ class ExtenderCore {
 
   FilterTracker tracker;
   NativeCounters counters;
+  LatencyHist filter_hist;
+  LatencyHist priorities_hist;
 
  private:
   std::shared_ptr<ClusterState> state_;

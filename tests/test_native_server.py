@@ -282,3 +282,29 @@ def test_nesting_bomb_rejected_not_crashed(native):
     assert core.json_roundtrip("[" * 200 + "1" + "]" * 200)
     with pytest.raises(RuntimeError):
         core.json_roundtrip("[" * 1000 + "1" + "]" * 1000)
+
+
+def test_native_latency_histograms_visible(native):
+    """the GIL-free fast path exports its own per-verb
+    latency histograms (C++-computed), surfaced on /metrics."""
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("h", core=10, memory=GiB))
+    with _client(fe) as c:
+        for _ in range(20):
+            c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+            c.post("/scheduler/priorities",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+    hists = fe.server.latency_histograms()
+    for verb in ("filter", "priorities"):
+        h = hists[verb]
+        assert h["count"] >= 20
+        assert h["sum_us"] > 0
+        assert sum(n for _, n in h["buckets"]) == h["count"]
+    # rendered on /metrics as a cumulative Prometheus histogram
+    st, _, body = fe.app.handle("GET", "/metrics", b"")
+    text = body.decode()
+    assert st == 200
+    assert 'egs_native_verb_latency_seconds_bucket{verb="filter"' in text
+    assert 'egs_native_verb_latency_seconds_count{verb="priorities"}' in text
+    assert 'le="+Inf"' in text
