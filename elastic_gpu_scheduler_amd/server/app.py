@@ -74,6 +74,8 @@ class ExtenderApp:
             ("GET", "/healthz"): self.healthz,
             ("GET", "/debug/stacks"): self.debug_stacks,
             ("GET", "/debug/profile"): self.debug_profile,
+            ("GET", "/debug/latency"): self.debug_latency,
+            ("GET", "/debug/heap"): self.debug_heap,
         }
 
     # ---- shared sync dispatcher -----------------------------------------
@@ -307,6 +309,62 @@ class ExtenderApp:
                          for stack, count in samples.most_common())
         return 200, None, (out + "\n").encode()
 
+    def debug_latency(self, body: bytes):
+        """Per-verb latency of BOTH request paths: the C++ fast path's
+        log2-us histograms (computed lock-free where the work happens —
+        the Python sampler cannot see those threads) and the Python-side
+        Prometheus histogram summaries. Closes the r1 gap of the GIL-free
+        95% of traffic being invisible to /debug/profile."""
+        out: Dict[str, Any] = {"native": None, "python_verbs": {}}
+        if self.native_server is not None:
+            hists = self.native_server.latency_histograms()
+            native = {}
+            for verb, h in hists.items():
+                buckets = [(le, n) for le, n in h["buckets"] if n]
+                native[verb] = {
+                    "count": h["count"],
+                    "mean_us": round(h["sum_us"] / h["count"], 1)
+                    if h["count"] else None,
+                    "buckets_us": buckets,
+                    "p50_us": _hist_quantile(h, 0.5),
+                    "p99_us": _hist_quantile(h, 0.99),
+                }
+            out["native"] = native
+        for m in metrics.REGISTRY.collect():
+            if m.name != "egs_verb_latency_seconds":
+                continue
+            for s in m.samples:
+                if s.name.endswith("_count"):
+                    verb = s.labels.get("verb", "?")
+                    out["python_verbs"].setdefault(verb, {})["count"] = s.value
+                elif s.name.endswith("_sum"):
+                    verb = s.labels.get("verb", "?")
+                    out["python_verbs"].setdefault(verb, {})["sum_s"] = \
+                        round(s.value, 6)
+        return 200, out, None
+
+    def debug_heap(self, body: bytes):
+        """Python heap snapshot (analogue of /debug/pprof/heap,
+        pkg/routes/pprof.go): top allocation sites by size via tracemalloc.
+        First call starts tracing and returns a baseline notice; later
+        calls return the top-50 sites. C++-side allocations are not
+        tracked (use the native counters + RSS for those)."""
+        import tracemalloc
+
+        if not tracemalloc.is_tracing():
+            tracemalloc.start(10)
+            return 200, {"tracing": "started; call again for a snapshot"}, None
+        snap = tracemalloc.take_snapshot()
+        stats = snap.statistics("lineno")[:50]
+        import resource
+
+        return 200, {
+            "rss_kib": resource.getrusage(resource.RUSAGE_SELF).ru_maxrss,
+            "traced_current_kib": tracemalloc.get_traced_memory()[0] // 1024,
+            "top": [{"site": str(s.traceback), "kib": s.size // 1024,
+                     "blocks": s.count} for s in stats],
+        }, None
+
     def debug_stacks(self, body: bytes):
         frames = sys._current_frames()
         out = []
@@ -314,6 +372,20 @@ class ExtenderApp:
             out.append(f"--- thread {tid} ---")
             out.extend(line.rstrip() for line in traceback.format_stack(frame))
         return 200, None, ("\n".join(out) + "\n").encode()
+
+
+def _hist_quantile(h, q: float):
+    """Approximate quantile from a log2-us histogram (upper bucket bound)."""
+    total = h["count"]
+    if not total:
+        return None
+    target = total * q
+    cum = 0
+    for le_us, n in h["buckets"]:
+        cum += n
+        if cum >= target:
+            return le_us
+    return h["buckets"][-1][0]
 
 
 class _BadRequest(Exception):

@@ -308,3 +308,26 @@ def test_native_latency_histograms_visible(native):
     assert 'egs_native_verb_latency_seconds_bucket{verb="filter"' in text
     assert 'egs_native_verb_latency_seconds_count{verb="priorities"}' in text
     assert 'le="+Inf"' in text
+
+
+def test_debug_latency_and_heap_endpoints(native):
+    client, registry, fe = native
+    pod = client.create_pod(make_pod("d", core=10, memory=GiB))
+    with _client(fe) as c:
+        for _ in range(5):
+            c.post("/scheduler/filter",
+                   json={"pod": pod, "nodenames": ["node-a"]})
+        r = c.get("/debug/latency")
+        assert r.status_code == 200
+        d = r.json()
+        assert d["native"]["filter"]["count"] >= 5
+        assert d["native"]["filter"]["p50_us"] is not None
+        assert "filter" in d["python_verbs"] or d["python_verbs"] == {}
+        # heap: first call arms tracemalloc, second returns a snapshot
+        r1 = c.get("/debug/heap")
+        assert r1.status_code == 200
+        r2 = c.get("/debug/heap")
+        assert r2.status_code == 200
+        snap = r2.json()
+        assert snap["rss_kib"] > 0
+        assert isinstance(snap["top"], list) and snap["top"]
