@@ -382,3 +382,52 @@ def test_real_wire_bench_smoke():
     assert result["metric"] == "pods_scheduled_per_sec_real_wire"
     assert result["value"] > 0
     assert result["config"]["bind_retries"] == 0
+
+
+def test_controller_reconciles_over_real_wire(real_client, apiserver):
+    """The full reconcile loop against the strict apiserver: bound-pod
+    replay through the chunked watch, release on DELETE, and accounting
+    restored over a watch drop — the informer behavior the reference gets
+    from client-go (controller.go), exercised on the real wire."""
+    from elastic_gpu_scheduler_amd.controller.controller import Controller
+    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+
+    apiserver.seed_node(make_node("w1", cards=2))
+    registry = SchedulerRegistry(real_client)
+    ctrl = Controller(real_client, registry, workers=1, resync_seconds=3600)
+    ctrl.start()
+    try:
+        sch = registry.default
+        # a pod bound by ANOTHER scheduler replica appears via the watch
+        pod = make_pod("replayed", core=40, memory=16 * GiB)
+        pod["metadata"]["annotations"] = {
+            "elasticgpu.io/assumed": "true",
+            "elasticgpu.io/container-c0": "1",
+        }
+        pod["metadata"]["labels"] = {"elasticgpu.io/assumed": "true"}
+        pod["spec"]["nodeName"] = "w1"
+        apiserver.seed_pod(pod)
+
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            if sch.state.has_node("w1") and \
+                    any(d.core_avail == 60
+                        for d in sch.state.node_devices("w1")):
+                break
+            time.sleep(0.05)
+        devs = sch.state.node_devices("w1")
+        assert any(d.core_avail == 60 for d in devs), devs
+
+        # watch drop + release during the gap: resumption must deliver it
+        apiserver.drop_watches()
+        apiserver.seed_delete_pod("default", "replayed")
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            if all(d.core_avail == 100
+                   for d in sch.state.node_devices("w1")):
+                break
+            time.sleep(0.05)
+        assert all(d.core_avail == 100
+                   for d in sch.state.node_devices("w1"))
+    finally:
+        ctrl.stop()
