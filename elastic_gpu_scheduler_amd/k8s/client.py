@@ -90,6 +90,14 @@ class KubeClient:
         """Register a pod watch; returns an unsubscribe callable."""
         raise NotImplementedError
 
+    def watch_nodes(self, handler: WatchHandler) -> Callable[[], None]:
+        """Register a node watch (agent inventory republish, node
+        add/remove). The reference creates a node informer and never
+        consults it (controller.go:97-99 — dead code); here node events
+        drive immediate node-cache invalidation instead of waiting for
+        the periodic resync."""
+        raise NotImplementedError
+
 
 def _match_labels(obj: Dict[str, Any], selector: Dict[str, str]) -> bool:
     labels = obj.get("metadata", {}).get("labels", {}) or {}
@@ -109,6 +117,7 @@ class FakeKubeClient(KubeClient):
         self._leases: Dict[str, Dict[str, Any]] = {}
         self._rv = 0
         self._watchers: List[WatchHandler] = []
+        self._node_watchers: List[WatchHandler] = []
 
     # -- helpers --
     def _next_rv(self) -> str:
@@ -118,6 +127,10 @@ class FakeKubeClient(KubeClient):
     def _notify(self, event_type: str, pod: Pod) -> None:
         for h in list(self._watchers):
             h(event_type, _jcopy(pod))
+
+    def _notify_node(self, event_type: str, node: Node) -> None:
+        for h in list(self._node_watchers):
+            h(event_type, _jcopy(node))
 
     @staticmethod
     def _key(namespace: str, name: str) -> str:
@@ -131,7 +144,15 @@ class FakeKubeClient(KubeClient):
             meta.setdefault("uid", str(uuid.uuid4()))
             meta["resourceVersion"] = self._next_rv()
             self._nodes[meta["name"]] = node
-            return _jcopy(node)
+            out = _jcopy(node)
+        self._notify_node("ADDED", out)
+        return out
+
+    def delete_node(self, name: str) -> None:
+        with self._mu:
+            node = self._nodes.pop(name, None)
+        if node is not None:
+            self._notify_node("DELETED", _jcopy(node))
 
     def create_pod(self, pod: Pod) -> Pod:
         with self._mu:
@@ -235,7 +256,9 @@ class FakeKubeClient(KubeClient):
             node.setdefault("metadata", {}).setdefault("annotations", {}).update(
                 annotations)
             node["metadata"]["resourceVersion"] = self._next_rv()
-            return _jcopy(node)
+            out = _jcopy(node)
+        self._notify_node("MODIFIED", out)
+        return out
 
     def patch_node_allocatable(self, name: str,
                                allocatable: Dict[str, str]) -> Node:
@@ -247,7 +270,9 @@ class FakeKubeClient(KubeClient):
             status.setdefault("allocatable", {}).update(allocatable)
             status.setdefault("capacity", {}).update(allocatable)
             node["metadata"]["resourceVersion"] = self._next_rv()
-            return _jcopy(node)
+            out = _jcopy(node)
+        self._notify_node("MODIFIED", out)
+        return out
 
     def create_event(self, namespace: str, event: Dict[str, Any]) -> None:
         with self._mu:
@@ -297,6 +322,17 @@ class FakeKubeClient(KubeClient):
             with self._mu:
                 if handler in self._watchers:
                     self._watchers.remove(handler)
+
+        return unsubscribe
+
+    def watch_nodes(self, handler: WatchHandler) -> Callable[[], None]:
+        with self._mu:
+            self._node_watchers.append(handler)
+
+        def unsubscribe() -> None:
+            with self._mu:
+                if handler in self._node_watchers:
+                    self._node_watchers.remove(handler)
 
         return unsubscribe
 
@@ -568,7 +604,8 @@ class RealKubeClient(KubeClient):
             f"/apis/coordination.k8s.io/v1/namespaces/{namespace}/leases/{name}",
             content=json.dumps(lease)))
 
-    def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
+    def _watch_resource(self, path: str,
+                        handler: WatchHandler) -> Callable[[], None]:
         """List+watch with resourceVersion resumption (informer semantics,
         which the reference gets from client-go — controller.go:24,106-116):
 
@@ -587,7 +624,7 @@ class RealKubeClient(KubeClient):
 
         def relist() -> Optional[str]:
             out = self._check(self._client.get(
-                "/api/v1/pods", params={"resourceVersion": "0"}))
+                path, params={"resourceVersion": "0"}))
             for p in out.get("items", []):
                 if stop.is_set():
                     return None
@@ -606,8 +643,7 @@ class RealKubeClient(KubeClient):
                     params = {"watch": "true", "allowWatchBookmarks": "true"}
                     if rv:
                         params["resourceVersion"] = rv
-                    with self._client.stream("GET", "/api/v1/pods",
-                                             params=params,
+                    with self._client.stream("GET", path, params=params,
                                              timeout=None) as resp:
                         if resp.status_code == 410:
                             rv = None  # too old: relist once, then resume
@@ -637,10 +673,18 @@ class RealKubeClient(KubeClient):
                         return
                     stop.wait(1.0)  # reconnect backoff; rv is kept -> resume
 
-        thread = threading.Thread(target=run, name="egs-watch", daemon=True)
+        thread = threading.Thread(target=run,
+                                  name=f"egs-watch-{path.rsplit('/', 1)[-1]}",
+                                  daemon=True)
         thread.start()
 
         def unsubscribe() -> None:
             stop.set()
 
         return unsubscribe
+
+    def watch_pods(self, handler: WatchHandler) -> Callable[[], None]:
+        return self._watch_resource("/api/v1/pods", handler)
+
+    def watch_nodes(self, handler: WatchHandler) -> Callable[[], None]:
+        return self._watch_resource("/api/v1/nodes", handler)
