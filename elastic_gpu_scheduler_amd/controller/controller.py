@@ -52,6 +52,23 @@ class Controller:
 
     def start(self) -> None:
         self._unsubscribe = self.client.watch_pods(self._on_event)
+        # Node watch: the agent republishing inventory/topology (or a node
+        # vanishing) invalidates the per-node allocator cache IMMEDIATELY
+        # instead of waiting for the periodic resync. The reference builds
+        # a node informer and never consults it (controller.go:97-99).
+        try:
+            self._unsubscribe_nodes = self.client.watch_nodes(
+                self._on_node_event)
+            # Baseline the RVs so the FIRST change after start already
+            # invalidates (watch backends without an initial replay would
+            # otherwise record-only the first event per node).
+            for n in self.client.list_nodes():
+                name = (n.get("metadata", {}) or {}).get("name")
+                if name:
+                    self._node_rv.setdefault(
+                        name, n.get("metadata", {}).get("resourceVersion"))
+        except NotImplementedError:
+            self._unsubscribe_nodes = None
         for i in range(self.workers):
             t = threading.Thread(target=self._worker, name=f"egs-sync-{i}",
                                  daemon=True)
@@ -66,6 +83,8 @@ class Controller:
         self._stop.set()
         if self._unsubscribe:
             self._unsubscribe()
+        if getattr(self, "_unsubscribe_nodes", None):
+            self._unsubscribe_nodes()
         for _ in self._threads:
             self._queue.put(None)
         for t in self._threads:
@@ -92,6 +111,30 @@ class Controller:
             pod = dict(pod)
             pod.setdefault("metadata", {})["_egs_deleted"] = True
         self._enqueue(key, pod)
+
+    def _on_node_event(self, event_type: str, node: Dict[str, Any]) -> None:
+        """Node ADDED/MODIFIED/DELETED: evict the allocator cache when the
+        node's resourceVersion moved past what we last saw (first sight is
+        record-only — invalidating a node we never tracked would churn the
+        cache at watch startup). Benign race with _resync_nodes: the worst
+        case is one extra invalidation, which only costs a lazy refill."""
+        name = (node.get("metadata", {}) or {}).get("name")
+        if not name:
+            return
+        schedulers = {id(s): s for s in self.registry.schedulers.values()}
+        if event_type == "DELETED":
+            self._node_rv.pop(name, None)
+            for sch in schedulers.values():
+                sch.invalidate_node(name)
+            return
+        rv = (node.get("metadata", {}) or {}).get("resourceVersion")
+        prev = self._node_rv.get(name)
+        self._node_rv[name] = rv
+        if prev is not None and rv != prev:
+            log.info("node %s changed (rv %s -> %s, watch); refreshing cache",
+                     name, prev, rv)
+            for sch in schedulers.values():
+                sch.invalidate_node(name)
 
     def _enqueue(self, key: str, pod: Dict[str, Any]) -> None:
         with self._pending_mu:

@@ -137,8 +137,8 @@ class StrictAPIServer:
         self._nodes: Dict[str, Dict[str, Any]] = {}
         self._leases: Dict[str, Dict[str, Any]] = {}
         self.events: List[Dict[str, Any]] = []
-        # watch log: (rv, type, snapshot) — compacted to event_retention
-        self._log: List[Tuple[int, str, Dict[str, Any]]] = []
+        # watch log: (rv, type, kind, snapshot) — compacted to event_retention
+        self._log: List[Tuple[int, str, str, Dict[str, Any]]] = []
         self._watch_gen = 0  # bumped by drop_watches()
         self.pause_watches = False  # test hook: refuse new watch streams
         self.request_counts: Dict[str, int] = {}
@@ -266,8 +266,9 @@ class StrictAPIServer:
         self._rv += 1
         return str(self._rv)
 
-    def _emit_locked(self, etype: str, pod: Dict[str, Any]) -> None:
-        self._log.append((self._rv, etype, json.loads(json.dumps(pod))))
+    def _emit_locked(self, etype: str, obj: Dict[str, Any],
+                     kind: str = "Pod") -> None:
+        self._log.append((self._rv, etype, kind, json.loads(json.dumps(obj))))
         if len(self._log) > self.event_retention:
             self._log = self._log[-self.event_retention:]
         self._watch_cv.notify_all()
@@ -287,6 +288,7 @@ class StrictAPIServer:
             node = json.loads(json.dumps(node))
             self._stamp_new_locked(node, None)
             self._nodes[node["metadata"]["name"]] = node
+            self._emit_locked("ADDED", node, kind="Node")
             return json.loads(json.dumps(node))
 
     def seed_pod(self, pod: Dict[str, Any]) -> Dict[str, Any]:
@@ -409,6 +411,9 @@ class StrictAPIServer:
                 self._serve_pod_list(h, params)
             return
         if method == "GET" and path == "/api/v1/nodes":
+            if params.get("watch") == "true":
+                self._serve_watch(h, params, kind="Node")
+                return
             with self._mu:
                 items = [json.loads(json.dumps(n))
                          for n in self._nodes.values()]
@@ -429,6 +434,7 @@ class StrictAPIServer:
                                     f'nodes "{name}" already exists')
                 self._stamp_new_locked(node, None)
                 self._nodes[name] = node
+                self._emit_locked("ADDED", node, kind="Node")
                 h._send_json(201, json.loads(json.dumps(node)))
             return
 
@@ -628,6 +634,7 @@ class StrictAPIServer:
                 raise _ApiError(404, "NotFound", f'nodes "{name}" not found')
             _deep_merge(node, patch)
             node["metadata"]["resourceVersion"] = self._next_rv_locked()
+            self._emit_locked("MODIFIED", node, kind="Node")
             h._send_json(200, json.loads(json.dumps(node)))
 
     # leases --------------------------------------------------------------
@@ -681,7 +688,8 @@ class StrictAPIServer:
             h._send_json(200, json.loads(json.dumps(lease)))
 
     # watch ---------------------------------------------------------------
-    def _serve_watch(self, h, params: Dict[str, str]) -> None:
+    def _serve_watch(self, h, params: Dict[str, str],
+                     kind: str = "Pod") -> None:
         if self.pause_watches:
             raise _ApiError(503, "ServiceUnavailable",
                             "watch refused (test pause)")
@@ -740,13 +748,13 @@ class StrictAPIServer:
                     if gen != self._watch_gen:
                         break
                     pending = [(rv, etype, snapshot)
-                               for rv, etype, snapshot in self._log
-                               if rv > cursor]
+                               for rv, etype, k, snapshot in self._log
+                               if rv > cursor and k == kind]
                     if not pending:
                         self._watch_cv.wait(timeout=self.bookmark_interval)
                         pending = [(rv, etype, snapshot)
-                                   for rv, etype, snapshot in self._log
-                                   if rv > cursor]
+                                   for rv, etype, k, snapshot in self._log
+                                   if rv > cursor and k == kind]
                         if gen != self._watch_gen:
                             break
                     rv_now = self._rv

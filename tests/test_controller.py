@@ -217,3 +217,52 @@ def test_resync_refreshes_changed_node_inventory():
         assert any(d.core_avail == 75 for d in devs)
     finally:
         ctrl.stop()
+
+
+def test_node_watch_invalidates_cache_immediately():
+    """Agent republish (node annotation patch) must refresh the scheduler's
+    node cache via the NODE WATCH — not only at the next periodic resync.
+    (The reference creates a node informer and never consults it,
+    controller.go:97-99.)"""
+    import json as _json
+
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=40, memory=GiB))
+        sch.assume(["n1"], pod)
+        sch.bind("n1", pod)
+        assert ctrl.wait_idle()
+        assert sch.state.has_node("n1")
+        # the agent publishes a richer inventory: 2 cards instead of default
+        inv = {"cards": [{"index": 0, "core": 100,
+                          "memory_bytes": 288 * GiB},
+                         {"index": 1, "core": 100,
+                          "memory_bytes": 288 * GiB}]}
+        client.patch_node_annotations(
+            "n1", {"elasticgpu.io/gpu-inventory": _json.dumps(inv)})
+        # invalidation is synchronous with the fake's watch fan-out: the
+        # cache entry is gone; next use refills from the new annotation
+        assert wait_until(lambda: not sch.state.has_node("n1"))
+        ok, _ = sch.assume(["n1"], client.create_pod(
+            make_pod("p2", core=40, memory=GiB)))
+        assert ok == ["n1"]
+        devs = sch.state.node_devices("n1")
+        assert len(devs) == 2
+        # the bound pod was replayed into the refreshed cache
+        assert any(d.core_avail == 60 for d in devs)
+    finally:
+        ctrl.stop()
+
+
+def test_node_watch_delete_evicts():
+    client, registry, ctrl = make_stack()
+    try:
+        sch = registry.default
+        pod = client.create_pod(make_pod("p", core=40, memory=GiB))
+        sch.assume(["n1"], pod)
+        assert sch.state.has_node("n1")
+        client.delete_node("n1")
+        assert wait_until(lambda: not sch.state.has_node("n1"))
+    finally:
+        ctrl.stop()
