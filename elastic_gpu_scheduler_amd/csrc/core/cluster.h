@@ -173,13 +173,26 @@ class ClusterState {
       }
     }
     std::vector<int> verdicts(names.size(), static_cast<int>(AssumeVerdict::kUnknownNode));
-    auto task = [&](int i) {
-      if (!allocs[i]) return;
+    // Inline cache pass first: warm nodes answer with two map lookups, so
+    // only the nodes that genuinely need a placement search pay the
+    // thread-pool dispatch.
+    std::vector<int> pending;
+    for (size_t i = 0; i < names.size(); ++i) {
+      if (!allocs[i]) continue;
+      int v = allocs[i]->assume_cached(uid, req, distinct);
+      if (v >= 0)
+        verdicts[i] = v ? static_cast<int>(AssumeVerdict::kOk)
+                        : static_cast<int>(AssumeVerdict::kInfeasible);
+      else
+        pending.push_back(static_cast<int>(i));
+    }
+    auto task = [&](int pi) {
+      int i = pending[pi];
       verdicts[i] = allocs[i]->assume(uid, req, *rater_, distinct)
                         ? static_cast<int>(AssumeVerdict::kOk)
                         : static_cast<int>(AssumeVerdict::kInfeasible);
     };
-    run_fanout(names.size(), task);
+    run_fanout(pending.size(), task);
     return verdicts;
   }
 
@@ -195,11 +208,17 @@ class ClusterState {
       }
     }
     std::vector<double> scores(names.size(), kScoreMin);
-    auto task = [&](int i) {
-      if (!allocs[i]) return;
+    std::vector<int> pending;
+    for (size_t i = 0; i < names.size(); ++i) {
+      if (!allocs[i]) continue;
+      if (!allocs[i]->score_cached(uid, req, distinct, &scores[i]))
+        pending.push_back(static_cast<int>(i));
+    }
+    auto task = [&](int pi) {
+      int i = pending[pi];
       scores[i] = allocs[i]->score(uid, req, *rater_, distinct);
     };
-    run_fanout(names.size(), task);
+    run_fanout(pending.size(), task);
     return scores;
   }
 

@@ -68,6 +68,43 @@ class NodeAllocator {
     return true;
   }
 
+  // Cache-only assume: 1 = feasible, 0 = infeasible, -1 = needs a search.
+  // The filter fan-out answers warm nodes inline with this (a mutex + two
+  // map lookups) and pays thread-pool dispatch only for the misses — with
+  // the shape cache, pool dispatch over 256 trivially-answerable nodes
+  // costs more than the answers themselves.
+  int assume_cached(const std::string& uid, const GPURequest& req,
+                    bool distinct = false) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = assumed_.find(uid);
+    if (it != assumed_.end()) return 1;
+    auto sit = shape_cache_.find(shape_hash(req, distinct));
+    if (sit == shape_cache_.end() || sit->second.gen != gen_) return -1;
+    if (!sit->second.result.feasible) return 0;
+    assumed_[uid] = {sit->second.result.option, now()};
+    return 1;
+  }
+
+  // Cache-only score; returns false when a search is needed.
+  bool score_cached(const std::string& uid, const GPURequest& req,
+                    bool distinct, double* out) {
+    std::lock_guard<std::mutex> g(mu_);
+    auto it = assumed_.find(uid);
+    if (it != assumed_.end()) {
+      *out = it->second.option.score;
+      return true;
+    }
+    auto sit = shape_cache_.find(shape_hash(req, distinct));
+    if (sit == shape_cache_.end() || sit->second.gen != gen_) return false;
+    if (!sit->second.result.feasible) {
+      *out = kScoreMin;
+      return true;
+    }
+    assumed_[uid] = {sit->second.result.option, now()};
+    *out = sit->second.result.option.score;
+    return true;
+  }
+
   // Score for prioritize. A cache miss re-runs the search (never crashes on a
   // missing option). Returns kScoreMin when infeasible.
   double score(const std::string& uid, const GPURequest& req, const Rater& rater,
