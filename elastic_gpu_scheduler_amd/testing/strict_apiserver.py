@@ -765,7 +765,7 @@ class StrictAPIServer:
                         time.monotonic() - last_bookmark >= \
                         self.bookmark_interval:
                     chunk({"type": "BOOKMARK",
-                           "object": {"kind": "Pod", "apiVersion": "v1",
+                           "object": {"kind": kind, "apiVersion": "v1",
                                       "metadata": {"resourceVersion":
                                                    str(rv_now)}}})
                     cursor = max(cursor, rv_now)
