@@ -67,6 +67,58 @@ int main() {
       }
     }
   }
+
+  // ThreadPool::parallel_for teardown race: each iteration
+  // creates fresh done_mu/done_cv on the caller's stack; the caller must
+  // never outrun the last worker's notify (use-after-destroy). Hammer it
+  // with concurrent short fan-outs — exactly the shape that raced.
+  {
+    ThreadPool pool(4);
+    std::vector<std::thread> drivers;
+    for (int d = 0; d < 8; ++d) {
+      drivers.emplace_back([&pool] {
+        std::atomic<int> sum{0};
+        for (int it = 0; it < 500; ++it)
+          pool.parallel_for(3, [&](int i) { sum.fetch_add(i); });
+        (void)sum;
+      });
+    }
+    for (auto& t : drivers) t.join();
+  }
+
+  // Large-cluster fan-out: enough nodes that ClusterState::assume takes
+  // the POOL path (> kInlineFanout pending searches), with concurrent
+  // binds bumping generations so shape-cache hits and misses interleave.
+  {
+    ClusterState big("binpack", 0, 4);
+    std::vector<std::string> bignames;
+    for (int i = 0; i < 96; ++i) {
+      std::vector<Device> devs(8);
+      for (auto& d : devs) d.mem_total = d.mem_avail = 288 * GiB;
+      std::string n = "b" + std::to_string(i);
+      big.add_node(n, devs, {});
+      bignames.push_back(n);
+    }
+    std::vector<std::thread> ts;
+    for (int w = 0; w < 8; ++w) {
+      ts.emplace_back([&, w] {
+        GPURequest req{GPUUnit{0, 10 + (w % 3) * 10, 2 * GiB}};
+        for (int it = 0; it < 60; ++it) {
+          std::string uid =
+              "big" + std::to_string(w) + "-" + std::to_string(it);
+          big.assume(bignames, uid, req);
+          big.score(bignames, uid, req);
+          try {
+            big.allocate(bignames[(w * 31 + it) % bignames.size()], uid, req);
+          } catch (const std::exception&) {
+          }
+          if (it % 2) big.forget_pod(uid);
+        }
+      });
+    }
+    for (auto& t : ts) t.join();
+  }
+
   std::printf("stress ok\n");
   return errors.load() ? 1 : 0;
 }
