@@ -139,3 +139,112 @@ def test_native_json_codec_matches_python(payload):
 
     encoded = json.dumps(payload)
     assert json.loads(core.json_roundtrip(encoded)) == payload
+
+
+# ---------------------------------------------------------------------------
+# Stateful: random op sequences against a pure-Python accounting model.
+# A handful of repeating shapes FORCES shape-cache sharing across pods
+# (the r2 large-cluster memo); the model proves the cache can never leak
+# stale feasibility into a commit (the reference's shape-cache sharing
+# double-books, allocate.go:30-33).
+
+_SHAPES = [
+    [core.GPUUnit(0, 30, 16 * GiB)],
+    [core.GPUUnit(0, 60, 64 * GiB)],
+    [core.GPUUnit(gpu_count=1)],
+    [core.GPUUnit(gpu_count=2)],
+]
+
+
+def _model_feasible(model_devs, shape):
+    u = shape[0]
+    if u.gpu_count > 0:
+        free = sum(1 for c, m in model_devs
+                   if c == 100 and m == 288 * GiB)
+        return free >= u.gpu_count
+    return any(c >= u.core and m >= u.memory for c, m in model_devs)
+
+
+_op = st.tuples(st.sampled_from(["assume", "allocate", "forget"]),
+                st.integers(0, 15),   # pod id
+                st.integers(0, 1),    # node
+                st.integers(0, len(_SHAPES) - 1))
+
+
+@settings(max_examples=120, deadline=None)
+@given(ops=st.lists(_op, min_size=1, max_size=60))
+def test_cluster_state_matches_python_model(ops):
+    state = core.ClusterState("binpack", 0, 2)
+    nodes = ["m0", "m1"]
+    for n in nodes:
+        state.add_node(n, [core.Device() for _ in range(3)], [])
+    # model: node -> [(core_avail, mem_avail)]; pod -> (node, shape, cards)
+    model = {n: [(100, 288 * GiB)] * 3 for n in nodes}
+    committed = {}
+    assumed_ok = set()  # uids holding a cached feasible placement
+
+    def apply(node, cards, shape, sign):
+        u = shape[0]
+        for c in cards:
+            ca, ma = model[node][c]
+            if u.gpu_count > 0:
+                model[node][c] = ((100, 288 * GiB) if sign < 0 else (0, 0))
+            else:
+                model[node][c] = (ca - sign * u.core,
+                                  ma - sign * u.memory)
+
+    for op, pid, nid, sid in ops:
+        uid = f"pod-{pid}"
+        node = nodes[nid]
+        shape = _SHAPES[sid]
+        if op == "assume":
+            verdicts = state.assume([node], uid, shape)
+            if uid not in committed and uid not in assumed_ok:
+                # no cached per-uid placement exists and assume commits
+                # nothing, so the native verdict must MATCH the model
+                expected = _model_feasible(model[node], shape)
+                assert (verdicts[0] == 0) == expected, (
+                    verdicts, shape, model[node])
+            if verdicts[0] == 0:
+                assumed_ok.add(uid)
+        elif op == "allocate":
+            if uid in committed:
+                continue  # idempotent re-allocate returns old placement
+            try:
+                opt = state.allocate(node, uid, shape)
+            except RuntimeError:
+                # must ONLY fail when the model agrees it cannot fit
+                assert not _model_feasible(model[node], shape), (
+                    model[node], shape)
+                continue
+            cards = [c for a in opt.allocated for c in a]
+            u = shape[0]
+            # returned placement must fit the MODEL's pre-state
+            if u.gpu_count > 0:
+                assert len(set(cards)) == u.gpu_count
+                for c in cards:
+                    assert model[node][c] == (100, 288 * GiB), \
+                        "double-booked whole card"
+            else:
+                (ca, ma) = model[node][cards[0]]
+                assert ca >= u.core and ma >= u.memory, "over-committed"
+            apply(node, cards, shape, +1)
+            committed[uid] = (node, shape, cards)
+        else:  # forget
+            state.forget_pod(uid)
+            # forget routes via the pod->node map, which only exists after
+            # allocate: an assumed-but-never-bound pod keeps its per-node
+            # cached placement until the TTL sweep (harmless in production:
+            # pod UIDs are never reused with a different spec) — so the
+            # model's "uid holds a cached placement" flag stays set.
+            if uid in committed:
+                assumed_ok.discard(uid)
+                node_c, shape_c, cards = committed.pop(uid)
+                apply(node_c, cards, shape_c, -1)
+
+    # final accounting must match the model exactly
+    for n in nodes:
+        devs = state.node_devices(n)
+        for i, d in enumerate(devs):
+            assert (d.core_avail, d.mem_avail) == model[n][i], (
+                n, i, (d.core_avail, d.mem_avail), model[n][i])
