@@ -431,3 +431,94 @@ def test_controller_reconciles_over_real_wire(real_client, apiserver):
                    for d in sch.state.node_devices("w1"))
     finally:
         ctrl.stop()
+
+
+def test_warm_start_recovers_over_real_wire(pki, apiserver):
+    """Crash recovery against the real wire: a fresh scheduler process
+    (new registry, same cluster) rebuilds all accounting from the assumed
+    pods it lists from the apiserver — the reference's restart path
+    (scheduler.go:86-106), here with label-selected LIST over mTLS."""
+    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+    from elastic_gpu_scheduler_amd.server.app import make_app
+
+    cfg = kind_style_kubeconfig(pki, apiserver)
+    c1 = RealKubeClient.from_kubeconfig(cfg)
+    try:
+        apiserver.seed_node(make_node("w1", cards=2))
+        registry = SchedulerRegistry(c1)
+        app = make_app(registry)
+        pods = [apiserver.seed_pod(make_pod(f"wp{i}", core=30,
+                                            memory=16 * GiB))
+                for i in range(3)]
+        for p in pods:
+            app.handle("POST", "/scheduler/filter", json.dumps(
+                {"pod": p, "nodenames": ["w1"]}).encode())
+            st, _, body = app.handle("POST", "/scheduler/bind", json.dumps({
+                "podName": p["metadata"]["name"], "podNamespace": "default",
+                "podUID": p["metadata"]["uid"], "node": "w1"}).encode())
+            assert st == 200, body
+        used_before = [d.core_avail
+                       for d in registry.default.state.node_devices("w1")]
+    finally:
+        c1.close()
+
+    # "restart": brand-new client + registry; warm start from the apiserver
+    c2 = RealKubeClient.from_kubeconfig(cfg)
+    try:
+        fresh = SchedulerRegistry(c2)
+        fresh.default._ensure_node("w1")
+        used_after = [d.core_avail
+                      for d in fresh.default.state.node_devices("w1")]
+        assert sorted(used_after) == sorted(used_before), (
+            used_before, used_after)
+        # and the recovered accounting is load-bearing: a 4th pod that
+        # does not fit is rejected
+        # 2 whole cards needed, but one card carries the recovered pods
+        big = apiserver.seed_pod(make_pod("big", core=200))
+        st, _, body = app_filter(fresh, big)
+        assert st == 200
+        out = json.loads(body)
+        assert out["nodenames"] == [] and "w1" in out["failedNodes"]
+    finally:
+        c2.close()
+
+
+def app_filter(registry, pod):
+    from elastic_gpu_scheduler_amd.server.app import make_app
+
+    app = make_app(registry)
+    return app.handle("POST", "/scheduler/filter", json.dumps(
+        {"pod": pod, "nodenames": ["w1"]}).encode())
+
+
+def test_agent_publish_over_real_wire(pki, apiserver, monkeypatch):
+    """The node agent's publish path (annotation patch + allocatable
+    status patch) against the strict server, consumed back by the
+    scheduler's node model."""
+    from elastic_gpu_scheduler_amd.agent import inventory as inv
+    from elastic_gpu_scheduler_amd.agent import topology as topo
+    from elastic_gpu_scheduler_amd.agent.agent import NodeAgent
+    from elastic_gpu_scheduler_amd.k8s import objects as obj
+
+    cards = [{"index": i, "memory_bytes": 288 * GiB, "core": 100}
+             for i in range(4)]
+    monkeypatch.setattr(inv, "discover", lambda prefer="auto": cards)
+    monkeypatch.setattr(topo, "discover",
+                        lambda n, prefer="auto": topo.default_hive(n))
+    apiserver.seed_node({"metadata": {"name": "agent-node"}, "status": {}})
+    client = RealKubeClient.from_kubeconfig(
+        kind_style_kubeconfig(pki, apiserver))
+    try:
+        agent = NodeAgent("agent-node", client)
+        agent.publish()
+        node = client.get_node("agent-node")
+        devs = obj.node_devices(node)
+        assert len(devs) == 4
+        assert all(d.core_total == 100 for d in devs)
+        hops = obj.node_topology(node)
+        assert len(hops) == 4 and hops[0][1] == 1
+        alloc = node["status"]["allocatable"]
+        assert alloc["elasticgpu.io/gpu-core"] == "400"
+        assert alloc["amd.com/gpu"] == "4"
+    finally:
+        client.close()
