@@ -271,6 +271,8 @@ class BenchPipeline:
                 for i in range(self.args.batch)]
         if self.args.no_http:
             self._schedule_direct(pods, record_latency)
+        elif self.args.kube_sim:
+            self._schedule_kube_sim(pods, record_latency)
         else:
             self._schedule_http(pods, record_latency)
         self._verify_sample(pods)
@@ -341,6 +343,83 @@ class BenchPipeline:
         if record_latency:
             with self._lat_mu:
                 self.latencies.append(time.perf_counter() - t0)
+
+    def _schedule_kube_sim(self, pods, record_latency: bool):
+        """--kube-sim: ONE kube-scheduler's actual concurrency shape —
+        scheduleOne runs filter -> priorities -> select SEQUENTIALLY (one
+        pod at a time), while binds are dispatched asynchronously (the
+        real scheduler's bind goroutine). The default load generator
+        models N independent schedulers racing; this mode shows the
+        conflict rate a standard single-scheduler deployment sees."""
+        import concurrent.futures as cf
+        import queue as _queue
+
+        sched_conn = MiniHttpClient("127.0.0.1", self.port)
+        bind_conns: "_queue.Queue" = _queue.Queue()
+        for _ in range(4):
+            bind_conns.put(MiniHttpClient("127.0.0.1", self.port))
+        errors = []
+
+        def do_bind(pod, node, t0):
+            conn = bind_conns.get()
+            try:
+                for attempt in range(8):
+                    status, out = conn.post_json("/scheduler/bind", {
+                        "podName": pod["metadata"]["name"],
+                        "podNamespace": "default",
+                        "podUID": pod["metadata"]["uid"],
+                        "node": node})
+                    if status == 200:
+                        break
+                    with self._lat_mu:
+                        self.bind_retries += 1
+                    # requeue: refilter like the real scheduler would
+                    status, body = conn.post_json(
+                        "/scheduler/filter",
+                        {"pod": pod, "nodenames": self.node_names})
+                    ok = body.get("nodenames") or []
+                    if not ok:
+                        errors.append(RuntimeError(f"infeasible: {body}"))
+                        return
+                    status, prio = conn.post_json(
+                        "/scheduler/priorities",
+                        {"pod": pod, "nodenames": ok})
+                    top = max(e["score"] for e in prio)
+                    tied = [e["host"] for e in prio if e["score"] == top]
+                    node = tied[(hash(pod["metadata"]["uid"]) + attempt + 1)
+                                % len(tied)]
+                else:
+                    errors.append(RuntimeError("bind kept failing"))
+                    return
+                if record_latency:
+                    with self._lat_mu:
+                        self.latencies.append(time.perf_counter() - t0)
+            finally:
+                bind_conns.put(conn)
+
+        with cf.ThreadPoolExecutor(max_workers=4) as binder:
+            futures = []
+            for pod in pods:
+                t0 = time.perf_counter()
+                _, body = sched_conn.post_json(
+                    "/scheduler/filter",
+                    {"pod": pod, "nodenames": self.node_names})
+                ok = body.get("nodenames") or []
+                if not ok:
+                    raise RuntimeError(f"no feasible node: {body}")
+                _, prio = sched_conn.post_json(
+                    "/scheduler/priorities", {"pod": pod, "nodenames": ok})
+                top = max(e["score"] for e in prio)
+                tied = [e["host"] for e in prio if e["score"] == top]
+                best = tied[hash(pod["metadata"]["uid"]) % len(tied)]
+                futures.append(binder.submit(do_bind, pod, best, t0))
+            for f in futures:
+                f.result()
+        sched_conn.close()
+        while not bind_conns.empty():
+            bind_conns.get().close()
+        if errors:
+            raise errors[0]
 
     def _schedule_direct(self, pods, record_latency: bool):
         """--no-http: drive the handlers in-process (core profiling mode)."""
@@ -436,6 +515,9 @@ def main():
                    help="HTTP front end: native C++ (default) or uvicorn")
     p.add_argument("--no-http", action="store_true",
                    help="bypass TCP; drive handlers in-process")
+    p.add_argument("--kube-sim", action="store_true",
+                   help="model ONE kube-scheduler (sequential scheduleOne, "
+                        "async binds) instead of N racing schedulers")
     p.add_argument("--no-verify", action="store_true",
                    help="skip on-GPU placement stamping")
     p.add_argument("--self-profile", default="", metavar="PATH",
@@ -569,6 +651,7 @@ def main():
                     "native_stats": (pipe._native.stats()
                                      if pipe._native else None),
                     "concurrency": args.concurrency,
+                    "kube_sim": args.kube_sim,
                     "pod_pad_bytes": args.pod_pad_bytes,
                     "p50_filter_bind_ms": round(p50, 3) if p50 else None,
                     "p99_filter_bind_ms": round(p99, 3) if p99 else None,
