@@ -114,11 +114,16 @@ def build_cluster(n_nodes: int, cards: int, use_gpu_inventory: bool):
 # ---------------------------------------------------------------- pipeline
 
 class MiniHttpClient:
-    """Minimal blocking HTTP/1.1 keep-alive client over one socket."""
+    """Minimal blocking HTTP/1.1 keep-alive client over one socket
+    (optionally TLS: pass an ssl.SSLContext)."""
 
-    def __init__(self, host: str, port: int):
-        self.sock = socket.create_connection((host, port))
-        self.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+    def __init__(self, host: str, port: int, ssl_context=None):
+        raw = socket.create_connection((host, port))
+        raw.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        if ssl_context is not None:
+            self.sock = ssl_context.wrap_socket(raw, server_hostname=host)
+        else:
+            self.sock = raw
         self.buf = b""
 
     def close(self):
@@ -213,6 +218,8 @@ class BenchPipeline:
         self._server_thread = None
         self._native = None
         self.base_url = None
+        self._tls_dir = None
+        self._client_ssl = None
         if not args.no_http:
             self._start_server()
         if use_gpu and not args.no_verify:
@@ -223,14 +230,32 @@ class BenchPipeline:
             self.probe = None
 
     def _start_server(self):
+        tls_kw = {}
+        if self.args.tls:
+            # full pipeline over HTTPS: OpenSSL terminates inside the C++
+            # server; the load generator speaks TLS on every connection
+            import ssl
+            import tempfile
+
+            from elastic_gpu_scheduler_amd.testing import generate_pki
+
+            self._tls_dir = tempfile.TemporaryDirectory(prefix="egs-bench-tls-")
+            pki = generate_pki(self._tls_dir.name)
+            tls_kw = {"tls_cert": pki["server_crt"],
+                      "tls_key": pki["server_key"]}
+            self._client_ssl = ssl.create_default_context(
+                cafile=pki["ca_crt"])
         if self.args.server == "native":
             from elastic_gpu_scheduler_amd.server.native import NativeFrontend
 
-            self._native = NativeFrontend(self.app, host="127.0.0.1", port=0)
+            self._native = NativeFrontend(self.app, host="127.0.0.1", port=0,
+                                          **tls_kw)
             self._native.start()
             self.port = self._native.port
             self.base_url = f"http://127.0.0.1:{self.port}"
             return
+        if tls_kw:
+            raise SystemExit("--tls requires --server native in the bench")
         import uvicorn
 
         port = self._free_port()
@@ -293,7 +318,8 @@ class BenchPipeline:
         n_workers = min(self.args.concurrency, len(pods))
 
         def worker():
-            conn = MiniHttpClient("127.0.0.1", self.port)
+            conn = MiniHttpClient("127.0.0.1", self.port,
+                                  ssl_context=self._client_ssl)
             try:
                 while True:
                     try:
@@ -354,10 +380,12 @@ class BenchPipeline:
         import concurrent.futures as cf
         import queue as _queue
 
-        sched_conn = MiniHttpClient("127.0.0.1", self.port)
+        sched_conn = MiniHttpClient("127.0.0.1", self.port,
+                                    ssl_context=self._client_ssl)
         bind_conns: "_queue.Queue" = _queue.Queue()
         for _ in range(4):
-            bind_conns.put(MiniHttpClient("127.0.0.1", self.port))
+            bind_conns.put(MiniHttpClient("127.0.0.1", self.port,
+                                          ssl_context=self._client_ssl))
         errors = []
 
         def do_bind(pod, node, t0):
@@ -483,6 +511,8 @@ class BenchPipeline:
 
     def close(self):
         self.controller.stop()
+        if self._tls_dir is not None:
+            self._tls_dir.cleanup()
         if self._native is not None:
             self._native.stop()
         if self._server is not None:
@@ -515,6 +545,9 @@ def main():
                    help="HTTP front end: native C++ (default) or uvicorn")
     p.add_argument("--no-http", action="store_true",
                    help="bypass TCP; drive handlers in-process")
+    p.add_argument("--tls", action="store_true",
+                   help="serve and drive the full pipeline over HTTPS "
+                        "(native front end, self-signed PKI)")
     p.add_argument("--kube-sim", action="store_true",
                    help="model ONE kube-scheduler (sequential scheduleOne, "
                         "async binds) instead of N racing schedulers")
@@ -652,6 +685,7 @@ def main():
                                      if pipe._native else None),
                     "concurrency": args.concurrency,
                     "kube_sim": args.kube_sim,
+                    "tls": args.tls,
                     "pod_pad_bytes": args.pod_pad_bytes,
                     "p50_filter_bind_ms": round(p50, 3) if p50 else None,
                     "p99_filter_bind_ms": round(p99, 3) if p99 else None,
