@@ -89,6 +89,7 @@ class NodeAllocator {
   bool score_cached(const std::string& uid, const GPURequest& req,
                     bool distinct, double* out) {
     std::lock_guard<std::mutex> g(mu_);
+    gc_assumed_locked();  // priorities-only traffic must not grow the map
     auto it = assumed_.find(uid);
     if (it != assumed_.end()) {
       *out = it->second.option.score;
