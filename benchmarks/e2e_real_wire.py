@@ -73,6 +73,25 @@ def run(args) -> dict:
 
     apiserver = _ApiHandle()
     node_names = [f"node-{i}" for i in range(args.nodes)]
+    # Everything below may fail (TLS, node creation, server start): the
+    # apiserver child must never outlive this process — guard the whole
+    # setup+run, tearing down whatever was built.
+    try:
+        return _run_with_apiserver(args, apiserver, node_names, tmp, pki)
+    finally:
+        apiserver.stop()
+        tmp.cleanup()
+
+
+def _run_with_apiserver(args, apiserver, node_names, tmp, pki):
+    from bench import MiniHttpClient
+
+    from elastic_gpu_scheduler_amd.controller.controller import Controller
+    from elastic_gpu_scheduler_amd.k8s.client import RealKubeClient
+    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+    from elastic_gpu_scheduler_amd.server.app import make_app
+    from elastic_gpu_scheduler_amd.server.native import NativeFrontend
+    from elastic_gpu_scheduler_amd.utils import types as t
 
     import base64
 
@@ -207,8 +226,6 @@ def run(args) -> dict:
         controller.stop()
         front.stop()
         client.close()
-        apiserver.stop()
-        tmp.cleanup()
 
     lat_ms = sorted(x * 1000 for x in latencies)
     return {
