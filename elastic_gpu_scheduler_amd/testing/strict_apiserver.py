@@ -142,6 +142,11 @@ class StrictAPIServer:
         self._watch_gen = 0  # bumped by drop_watches()
         self.pause_watches = False  # test hook: refuse new watch streams
         self.request_counts: Dict[str, int] = {}
+        # chaos hook: probability that a MUTATING request (POST/PUT/PATCH/
+        # DELETE, except watch) fails with a 500 BEFORE touching state —
+        # models a flaky apiserver for fault-injection e2e.
+        self.fault_rate = 0.0
+        self._fault_seq = 0
 
         outer = self
 
@@ -396,6 +401,15 @@ class StrictAPIServer:
         r"^/apis/coordination\.k8s\.io/v1/namespaces/([^/]+)/leases/([^/]+)$")
 
     def _route(self, h, method: str) -> None:
+        if self.fault_rate > 0 and method in ("POST", "PUT", "PATCH",
+                                              "DELETE"):
+            # deterministic pseudo-random sequence so runs are replayable
+            with self._mu:
+                self._fault_seq += 1
+                seq = self._fault_seq
+            if (seq * 2654435761 % 997) / 997.0 < self.fault_rate:
+                raise _ApiError(500, "InternalError",
+                                "etcdserver: request timed out (injected)")
         path, _, query = h.path.partition("?")
         params: Dict[str, str] = {}
         for part in query.split("&"):
