@@ -288,6 +288,17 @@ class NodeAllocator {
   }
 
   bool fits_locked(const GPURequest& req, const GPUOption& option) const {
+    // The option must STRUCTURALLY match the request (right card count per
+    // container) — a cached option from a different shape (possible only
+    // through API misuse; pod specs are immutable in k8s) must never
+    // under- or over-allocate silently. Mismatch -> caller re-searches.
+    if (option.allocated.size() < req.size()) return false;
+    for (size_t c = 0; c < req.size(); ++c) {
+      const GPUUnit& u = req[c];
+      size_t want = u.whole_cards() ? static_cast<size_t>(u.gpu_count)
+                                    : (u.needs_gpu() ? 1 : 0);
+      if (option.allocated[c].size() != want) return false;
+    }
     std::vector<Device> copy = devices_;
     for (size_t c = 0; c < option.allocated.size() && c < req.size(); ++c) {
       const GPUUnit& u = req[c];
