@@ -169,10 +169,15 @@ def main(argv=None) -> int:
     else:
         import uvicorn
 
+        import ssl as _ssl
+
         config = uvicorn.Config(app, host=args.host, port=args.port,
                                 log_level=args.log_level, access_log=False,
                                 ssl_certfile=args.tls_cert or None,
-                                ssl_keyfile=args.tls_key or None)
+                                ssl_keyfile=args.tls_key or None,
+                                ssl_ca_certs=args.tls_client_ca or None,
+                                ssl_cert_reqs=_ssl.CERT_REQUIRED
+                                if args.tls_client_ca else _ssl.CERT_NONE)
         server = uvicorn.Server(config)
 
         def watch_stop():
