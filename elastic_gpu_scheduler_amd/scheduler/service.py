@@ -56,6 +56,21 @@ class GPUUnitScheduler:
         # fast-path membership cache over state's node map (adds only; a
         # removed node falls back to the authoritative native check)
         self._known_nodes: set = set()
+        # Binds in flight per node + deferred invalidations: evicting a
+        # node between a bind's in-memory allocate and its annotation
+        # write would lose the reservation — the refill replays only
+        # apiserver-VISIBLE pods, so a concurrent bind on the refilled
+        # state could double-book the same cards. Invalidation therefore
+        # defers until the node's last in-flight bind completes.
+        self._bind_mu = threading.Lock()
+        self._binds_inflight: Dict[str, int] = {}
+        self._dirty_nodes: set = set()
+        # Serialises evict/refill transitions: two threads refilling the
+        # same node would replace each other's allocator MID-REPLAY and
+        # drop accounting for already-replayed pods. Never taken on the
+        # warm path (known-node check first). Lock order: _bind_mu may be
+        # held when taking _fill_mu, never the reverse.
+        self._fill_mu = threading.Lock()
         # events are emitted asynchronously: an apiserver event write must
         # never sit on the bind critical path
         self._event_q: "queue.Queue" = queue.Queue(maxsize=4096)
@@ -91,39 +106,70 @@ class GPUUnitScheduler:
         (reference getNodeInfo, scheduler.go:62-84)."""
         if node_name in self._known_nodes:
             return True
-        if self.state.has_node(node_name):
-            self._known_nodes.add(node_name)
-            return True
-        try:
-            node = self.client.get_node(node_name)
-        except NotFoundError:
-            return False
-        except Exception:
-            log.exception("get node %s failed", node_name)
-            return False
-        devices = obj.node_devices(node, self.bare_unit)
-        if not devices:
-            return False
-        topo = obj.node_topology(node)
-        self.state.add_node(node_name, devices, topo)
-        self._known_nodes.add(node_name)
-        if replay:
+        with self._fill_mu:
+            if self.state.has_node(node_name):  # lost the refill race: done
+                self._known_nodes.add(node_name)
+                return True
             try:
-                pods = self.client.list_pods(
-                    label_selector={t.EGPU_ASSUMED: "true"},
-                    field_selector={"spec.nodeName": node_name})
+                node = self.client.get_node(node_name)
+            except NotFoundError:
+                return False
             except Exception:
-                pods = []
-            for pod in pods:
-                if not obj.is_completed_pod(pod):
-                    self._replay_pod(node_name, pod)
+                log.exception("get node %s failed", node_name)
+                return False
+            devices = obj.node_devices(node, self.bare_unit)
+            if not devices:
+                return False
+            topo = obj.node_topology(node)
+            self.state.add_node(node_name, devices, topo)
+            if replay:
+                try:
+                    pods = self.client.list_pods(
+                        label_selector={t.EGPU_ASSUMED: "true"},
+                        field_selector={"spec.nodeName": node_name})
+                except Exception:
+                    pods = []
+                for pod in pods:
+                    if not obj.is_completed_pod(pod):
+                        self._replay_pod(node_name, pod)
+            self._known_nodes.add(node_name)
         return True
 
     def invalidate_node(self, node_name: str) -> None:
         """Evict a node from the cache so fresh inventory/topology (e.g. the
-        agent republishing annotations) is re-read on next use."""
-        self._known_nodes.discard(node_name)
-        self.state.remove_node(node_name)
+        agent republishing annotations) is re-read on next use. If a bind
+        is mid-flight on this node the eviction is DEFERRED to the last
+        bind's completion — evicting now would drop its not-yet-annotated
+        reservation and let the refilled cache double-book those cards
+        (found by tests/test_concurrency.py invalidation-churn)."""
+        with self._bind_mu:
+            if self._binds_inflight.get(node_name, 0) > 0:
+                self._dirty_nodes.add(node_name)
+                return
+            # evict UNDER the lock: a bind entering concurrently must only
+            # proceed after the eviction, so its _ensure_node refills first
+            self._evict_node(node_name)
+
+    def _evict_node(self, node_name: str) -> None:
+        with self._fill_mu:  # atomic vs any in-progress refill
+            self._known_nodes.discard(node_name)
+            self.state.remove_node(node_name)
+
+    def _bind_enter(self, node_name: str) -> None:
+        with self._bind_mu:
+            self._binds_inflight[node_name] = \
+                self._binds_inflight.get(node_name, 0) + 1
+
+    def _bind_exit(self, node_name: str) -> None:
+        with self._bind_mu:
+            n = self._binds_inflight.get(node_name, 1) - 1
+            if n <= 0:
+                self._binds_inflight.pop(node_name, None)
+                if node_name in self._dirty_nodes:
+                    self._dirty_nodes.discard(node_name)
+                    self._evict_node(node_name)  # under the lock, see above
+            else:
+                self._binds_inflight[node_name] = n
 
     def _replay_pod(self, node_name: str, pod: Dict[str, Any]) -> None:
         allocated = obj.parse_allocation(pod)
@@ -172,18 +218,26 @@ class GPUUnitScheduler:
         allocation if the apiserver writes fail."""
         uid = obj.pod_uid(pod)
         req = obj.pod_gpu_request(pod, self.bare_unit)
-        if not self._ensure_node(node_name):
-            raise BindError(f"unknown or GPU-less node {node_name}")
+        # Inside the enter/exit window any invalidate_node defers: the
+        # in-memory allocation below is invisible to the apiserver until
+        # _write_bind lands, and an eviction+refill in between would let a
+        # concurrent bind double-book these cards.
+        self._bind_enter(node_name)
         try:
-            option = self.state.allocate(node_name, uid, req,
-                                         obj.wants_container_spread(pod))
-        except RuntimeError as exc:
-            raise BindError(str(exc)) from exc
-        try:
-            self._write_bind(node_name, pod, option)
-        except Exception:
-            self.state.forget_pod(uid)
-            raise
+            if not self._ensure_node(node_name):
+                raise BindError(f"unknown or GPU-less node {node_name}")
+            try:
+                option = self.state.allocate(node_name, uid, req,
+                                             obj.wants_container_spread(pod))
+            except RuntimeError as exc:
+                raise BindError(str(exc)) from exc
+            try:
+                self._write_bind(node_name, pod, option)
+            except Exception:
+                self.state.forget_pod(uid)
+                raise
+        finally:
+            self._bind_exit(node_name)
         self._emit_event(pod, "Scheduled",
                          f"placed on {node_name} gpus "
                          f"{[list(a) for a in option.allocated]} "

@@ -163,3 +163,98 @@ def test_soak_assumed_cache_stays_bounded():
         # hold the line instead (8192 per node).
         assert c.node_assumed_count(n) <= 8192
         assert c.node_pods(n) == []
+
+
+def test_invalidation_churn_under_load():
+    """Agent republish (node-cache invalidation) racing live scheduling:
+    binds and invalidations interleave for a while; afterwards the
+    scheduler's accounting must exactly match the apiserver ground truth
+    (every invalidation triggers a lazy refill + assumed-pod replay)."""
+    import json as _json
+    import threading
+
+    from elastic_gpu_scheduler_amd.k8s import objects as obj
+    from elastic_gpu_scheduler_amd.k8s.client import FakeKubeClient
+    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
+    from tests.conftest import make_node, make_pod
+
+    GiB = 1024**3
+    client = FakeKubeClient()
+    inv = {"cards": [{"index": i, "core": 100, "memory_bytes": 288 * GiB}
+                     for i in range(8)]}
+    client.add_node(make_node(
+        "n1", annotations={"elasticgpu.io/gpu-inventory": _json.dumps(inv)}))
+    registry = SchedulerRegistry(client)
+    sch = registry.default
+
+    stop = threading.Event()
+    bound = []
+    bound_mu = threading.Lock()
+    errors = []
+
+    from elastic_gpu_scheduler_amd.scheduler.service import BindError
+
+    def scheduler_loop(wid):
+        # 30 pods/worker: 120 x 5 core of the 160-pod capacity, so the
+        # final probe still fits
+        for i in range(30):
+            if stop.is_set():
+                return
+            pod = client.create_pod(
+                make_pod(f"churn-{wid}-{i}", core=5, memory=GiB))
+            # kube-scheduler semantics: a failed bind requeues the pod
+            for attempt in range(50):
+                try:
+                    ok, _ = sch.assume(["n1"], pod)
+                    if not ok:
+                        break
+                    sch.bind("n1", client.get_pod(
+                        "default", pod["metadata"]["name"]))
+                    with bound_mu:
+                        bound.append(pod["metadata"]["name"])
+                    break
+                except BindError:
+                    continue  # transient (invalidation race): requeue
+                except Exception as exc:  # noqa: BLE001
+                    errors.append(exc)
+                    break
+
+    def invalidator_loop():
+        while not stop.is_set():
+            # the agent republishing bumps the node RV; the controller
+            # would call invalidate_node — do it directly here (far more
+            # often than any real agent would)
+            sch.invalidate_node("n1")
+            _time.sleep(0.001)
+
+    import time as _time
+
+    workers = [threading.Thread(target=scheduler_loop, args=(w,))
+               for w in range(4)]
+    inval = threading.Thread(target=invalidator_loop)
+    for t in workers:
+        t.start()
+    inval.start()
+    for t in workers:
+        t.join()
+    stop.set()
+    inval.join()
+    assert not errors, errors[:3]
+    assert len(bound) == 120, f"only {len(bound)} of 120 bound"
+
+    # ground truth from the apiserver: every bound pod's annotations
+    truth = [0] * 8
+    for name in bound:
+        pod = client.get_pod("default", name)
+        assert obj.is_assumed(pod)
+        alloc = obj.parse_allocation(pod)
+        for cards in alloc:
+            for c in cards:
+                truth[c] += 5
+    sch.invalidate_node("n1")  # final refill replays everything
+    ok, _ = sch.assume(["n1"], client.create_pod(
+        make_pod("probe", core=5, memory=GiB)))
+    assert ok == ["n1"]
+    devs = sch.state.node_devices("n1")
+    used = [d.core_total - d.core_avail for d in devs]
+    assert used == truth, (used, truth)
