@@ -37,15 +37,7 @@ GiB = 1024**3
 
 
 def run(args) -> dict:
-    from bench import MiniHttpClient
-
-    from elastic_gpu_scheduler_amd.controller.controller import Controller
-    from elastic_gpu_scheduler_amd.k8s.client import RealKubeClient
-    from elastic_gpu_scheduler_amd.scheduler.service import SchedulerRegistry
-    from elastic_gpu_scheduler_amd.server.app import make_app
-    from elastic_gpu_scheduler_amd.server.native import NativeFrontend
     from elastic_gpu_scheduler_amd.testing import generate_pki
-    from elastic_gpu_scheduler_amd.utils import types as t
 
     tmp = tempfile.TemporaryDirectory(prefix="egs-e2e-pki-")
     pki = generate_pki(tmp.name)
@@ -77,13 +69,13 @@ def run(args) -> dict:
     # apiserver child must never outlive this process — guard the whole
     # setup+run, tearing down whatever was built.
     try:
-        return _run_with_apiserver(args, apiserver, node_names, tmp, pki)
+        return _run_with_apiserver(args, apiserver, node_names, pki)
     finally:
         apiserver.stop()
         tmp.cleanup()
 
 
-def _run_with_apiserver(args, apiserver, node_names, tmp, pki):
+def _run_with_apiserver(args, apiserver, node_names, pki):
     from bench import MiniHttpClient
 
     from elastic_gpu_scheduler_amd.controller.controller import Controller
