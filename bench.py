@@ -598,6 +598,41 @@ def main():
         with _quiet_stdout():
             dist.init_process_group(backend="nccl" if use_gpu else "gloo")
 
+    # Opportunistic multi-GPU topology capture (the
+    # >=2-card branches of probe.hip had never seen hardware): when this
+    # bench lands on a multi-GPU box (the driver's 8-GPU scaling run),
+    # rank 0 measures the REAL xGMI hop matrix + p2p bandwidths once,
+    # before the timed region, in a SUBPROCESS with a hard timeout so a
+    # wedged p2p path can never hang the bench itself.
+    xgmi_measured = None
+    if use_gpu and rank == 0 and torch.cuda.device_count() > 1:
+        import subprocess
+        import sys as _sys
+
+        try:
+            probe_out = subprocess.run(
+                [_sys.executable, "-c",
+                 "import json\n"
+                 "from elastic_gpu_scheduler_amd.agent.agent import NodeAgent\n"
+                 "t = NodeAgent('bench').measured_topology(mib=8, iters=2)\n"
+                 "print(json.dumps(t))"],
+                capture_output=True, text=True, timeout=150,
+                cwd=os.path.dirname(os.path.abspath(__file__)))
+            if probe_out.returncode == 0:
+                topo = json.loads(probe_out.stdout.strip().splitlines()[-1])
+                bw = topo.get("bandwidth_gbps") or []
+                xgmi_measured = {
+                    "hops": topo.get("hops"),
+                    "bandwidth_gbps": [[round(x) for x in row]
+                                       for row in bw],
+                }
+            else:
+                print(f"xgmi capture failed: {probe_out.stderr[-400:]}",
+                      file=__import__("sys").stderr)
+        except Exception as exc:  # never fail the bench for the probe
+            print(f"xgmi capture skipped: {exc}",
+                  file=__import__("sys").stderr)
+
     device_index = local_rank % max(
         torch.cuda.device_count(), 1) if use_gpu else 0
     pipe = BenchPipeline(rank, args, device_index, use_gpu)
@@ -691,6 +726,7 @@ def main():
                     "p99_filter_bind_ms": round(p99, 3) if p99 else None,
                     "verify_on_device": pipe.probe is not None,
                     "bind_retries": pipe.bind_retries,
+                    "xgmi_measured": xgmi_measured,
                 },
             }
             print(json.dumps(result), flush=True)
