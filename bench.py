@@ -568,6 +568,11 @@ def main():
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world_size > 1 and args.filter_threads == 0:
+        # one scheduler shard per GPU shares the host's CPUs: size each
+        # shard's fan-out pool to its fair share instead of ncpu threads
+        # PER RANK (8 ranks x 256 threads would thrash a 256-core box)
+        args.filter_threads = max(4, (os.cpu_count() or 8) // world_size)
     use_gpu = torch.cuda.is_available()
     distributed = world_size > 1
 
