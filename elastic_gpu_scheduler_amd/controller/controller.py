@@ -11,9 +11,10 @@ AddPod for running assigned pods, ForgetPod for completed/deleted ones
     machinery; resync is an explicit periodic relist (reference uses a 30 s
     shared-informer resync, controller.go:24);
   * the node informer the reference creates but never consults
-    (controller.go:97-99) is replaced by node-cache invalidation: node
-    deletions/changes evict the per-node allocator so inventory updates
-    (e.g. the agent republishing topology) take effect.
+    (controller.go:97-99) is ALIVE here: a node watch invalidates the
+    per-node allocator cache immediately on inventory/topology republish
+    or node deletion (plus a relist fallback in the periodic resync), so
+    agent updates take effect without waiting out the resync period.
 """
 from __future__ import annotations
 
