@@ -1,7 +1,7 @@
 # Build/test entry points (reference Makefile:1-13 analogue).
 TAG ?= elastic-gpu-scheduler-amd:latest
 
-.PHONY: native test test-gpu bench image clean
+.PHONY: native test test-gpu bench e2e image clean
 
 native:
 	python3 build_native.py
@@ -14,6 +14,12 @@ test-gpu: native
 
 bench: native
 	python3 bench.py --steps 20 --warmup 3
+
+# Wire-strict end-to-end: mTLS control plane + native HTTPS + the
+# real-wire config-1 measurement (apiserver as its own process).
+e2e: native
+	python3 -m pytest tests/test_strict_apiserver_e2e.py tests/test_native_tls.py -q
+	python3 benchmarks/e2e_real_wire.py --steps 2 --warmup 1 --batch 50
 
 image:
 	docker build -t $(TAG) .
