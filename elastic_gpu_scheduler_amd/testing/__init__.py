@@ -7,8 +7,19 @@ typed Status errors, resourceVersion optimistic concurrency, watch framing
 with bookmarks and 410 Gone, mTLS/Bearer auth) instead of the permissive
 in-memory fake. Wire-format bugs that only a live apiserver would catch —
 like r1's unix-float Lease renewTime — fail loudly here.
+
+Lazy exports: `python -m elastic_gpu_scheduler_amd.testing.strict_apiserver`
+runs the server as its own process (bench use), and an eager import here
+would double-import the module under runpy.
 """
-from elastic_gpu_scheduler_amd.testing.strict_apiserver import (  # noqa: F401
-    StrictAPIServer,
-    generate_pki,
-)
+from __future__ import annotations
+
+__all__ = ["StrictAPIServer", "generate_pki"]
+
+
+def __getattr__(name: str):
+    if name in __all__:
+        from elastic_gpu_scheduler_amd.testing import strict_apiserver as m
+
+        return getattr(m, name)
+    raise AttributeError(name)
