@@ -18,6 +18,7 @@
 #include <algorithm>
 #include <chrono>
 #include <cstdint>
+#include <deque>
 #include <mutex>
 #include <optional>
 #include <stdexcept>
@@ -64,7 +65,7 @@ class NodeAllocator {
     if (it != assumed_.end()) return true;
     const SearchResult& res = shape_search_locked(req, rater, distinct);
     if (!res.feasible) return false;
-    assumed_[uid] = {res.option, now()};
+    remember_assumed_locked(uid, res.option);
     return true;
   }
 
@@ -81,7 +82,7 @@ class NodeAllocator {
     auto sit = shape_cache_.find(shape_hash(req, distinct));
     if (sit == shape_cache_.end() || sit->second.gen != gen_) return -1;
     if (!sit->second.result.feasible) return 0;
-    assumed_[uid] = {sit->second.result.option, now()};
+    remember_assumed_locked(uid, sit->second.result.option);
     return 1;
   }
 
@@ -101,7 +102,7 @@ class NodeAllocator {
       *out = kScoreMin;
       return true;
     }
-    assumed_[uid] = {sit->second.result.option, now()};
+    remember_assumed_locked(uid, sit->second.result.option);
     *out = sit->second.result.option.score;
     return true;
   }
@@ -115,7 +116,7 @@ class NodeAllocator {
     if (it != assumed_.end()) return it->second.option.score;
     const SearchResult& res = shape_search_locked(req, rater, distinct);
     if (!res.feasible) return kScoreMin;
-    assumed_[uid] = {res.option, now()};
+    remember_assumed_locked(uid, res.option);
     return res.option.score;
   }
 
@@ -353,34 +354,32 @@ class NodeAllocator {
     }
   }
 
-  void gc_assumed_locked() {
-    // A pod assumes on EVERY filtered node but binds on one, so entries for
-    // the losing nodes linger. Sweep expired entries once the map is large,
-    // and hard-cap the map by evicting the oldest half — sustained
-    // scheduling must not grow memory without bound. The sweep itself is
-    // O(map): gate it by time so steady-state assumes stay O(1) (an
-    // every-call sweep measurably throttled sustained throughput).
-    if (assumed_.size() < 1024) return;  // amortise: only sweep when large
+  // Record an assumed placement and keep the map bounded. A pod assumes
+  // on EVERY filtered node but binds on one, so entries for the losing
+  // nodes linger; at 256-node scale the r1 sweep (O(map) scan +
+  // nth_element eviction whenever the map hit its cap) showed up as
+  // periodic p99 spikes in sustained soaks. Entries are never refreshed,
+  // so INSERTION ORDER == age order: an insertion-order queue makes both
+  // TTL expiry and cap eviction O(1) amortised per operation.
+  void remember_assumed_locked(const std::string& uid, const GPUOption& opt) {
     auto t = now();
-    if (assumed_.size() < 8192 && t - last_gc_ < std::chrono::seconds(5))
-      return;
-    last_gc_ = t;
+    assumed_[uid] = {opt, t};
+    assume_order_.emplace_back(t, uid);
+  }
+
+  void gc_assumed_locked() {
+    constexpr size_t kSoftCap = 8192;
     auto cutoff = now() - kAssumeTTL;
-    for (auto it = assumed_.begin(); it != assumed_.end();) {
-      if (it->second.at < cutoff)
-        it = assumed_.erase(it);
-      else
-        ++it;
+    while (!assume_order_.empty()) {
+      bool overflow = assumed_.size() > kSoftCap;
+      const auto& [t, uid] = assume_order_.front();
+      if (t >= cutoff && !overflow) break;
+      auto it = assumed_.find(uid);
+      // the timestamp guard keeps a RE-assumed uid (erased at allocate,
+      // assumed again later) from being evicted by its stale record
+      if (it != assumed_.end() && it->second.at == t) assumed_.erase(it);
+      assume_order_.pop_front();
     }
-    constexpr size_t kHardCap = 8192;
-    if (assumed_.size() <= kHardCap) return;
-    std::vector<std::pair<Clock::time_point, std::string>> order;
-    order.reserve(assumed_.size());
-    for (const auto& [uid, e] : assumed_) order.emplace_back(e.at, uid);
-    std::nth_element(order.begin(), order.begin() + order.size() / 2,
-                     order.end());
-    for (size_t i = 0; i < order.size() / 2; ++i)
-      assumed_.erase(order[i].second);
   }
 
   struct ShapeEntry {
@@ -388,7 +387,6 @@ class NodeAllocator {
     SearchResult result;
   };
 
-  Clock::time_point last_gc_ = Clock::time_point::min();
   uint64_t gen_ = 1;  // device-state generation (shape-cache validity)
   std::unordered_map<uint64_t, ShapeEntry> shape_cache_;
   std::string name_;
@@ -397,6 +395,8 @@ class NodeAllocator {
   double topology_weight_ = kDefaultTopologyWeight;
   std::mutex mu_;
   std::unordered_map<std::string, Assumed> assumed_;      // uid -> pending placement
+  // insertion-order records for O(1) TTL/cap eviction (see gc_assumed_locked)
+  std::deque<std::pair<Clock::time_point, std::string>> assume_order_;
   std::unordered_map<std::string, GPUOption> pods_;       // uid -> committed placement
   std::unordered_map<std::string, GPURequest> requests_;  // uid -> demand (for cancel)
 };

@@ -203,7 +203,9 @@ def test_invalidation_churn_under_load():
             pod = client.create_pod(
                 make_pod(f"churn-{wid}-{i}", core=5, memory=GiB))
             # kube-scheduler semantics: a failed bind requeues the pod
-            for attempt in range(50):
+            # (generous budget: under xdist CPU contention the 1 kHz
+            # invalidator can win many rounds in a row)
+            for attempt in range(500):
                 try:
                     ok, _ = sch.assume(["n1"], pod)
                     if not ok:
