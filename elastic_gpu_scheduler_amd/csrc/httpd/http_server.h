@@ -102,22 +102,46 @@ class HttpServer {
             nullptr);
       }
     }
-    listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+    // Dual-stack: an IPv6 literal (or "::") binds AF_INET6 with
+    // V6ONLY off, so IPv6-first clusters work; anything else is IPv4.
+    bool v6 = host.find(':') != std::string::npos;
+    listen_fd_ = ::socket(v6 ? AF_INET6 : AF_INET, SOCK_STREAM, 0);
     if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
     int one = 1;
     setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
-    sockaddr_in addr{};
-    addr.sin_family = AF_INET;
-    addr.sin_port = htons(static_cast<uint16_t>(port));
-    if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1)
-      addr.sin_addr.s_addr = INADDR_ANY;
-    if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)) < 0) {
-      ::close(listen_fd_);
-      throw std::runtime_error("bind() failed on port " + std::to_string(port));
+    if (v6) {
+      int zero = 0;
+      setsockopt(listen_fd_, IPPROTO_IPV6, IPV6_V6ONLY, &zero, sizeof(zero));
+      sockaddr_in6 addr6{};
+      addr6.sin6_family = AF_INET6;
+      addr6.sin6_port = htons(static_cast<uint16_t>(port));
+      if (inet_pton(AF_INET6, host.c_str(), &addr6.sin6_addr) != 1)
+        addr6.sin6_addr = in6addr_any;
+      if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr6),
+                 sizeof(addr6)) < 0) {
+        ::close(listen_fd_);
+        throw std::runtime_error("bind() failed on port " +
+                                 std::to_string(port));
+      }
+      socklen_t len6 = sizeof(addr6);
+      getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&addr6), &len6);
+      port_ = ntohs(addr6.sin6_port);
+    } else {
+      sockaddr_in addr{};
+      addr.sin_family = AF_INET;
+      addr.sin_port = htons(static_cast<uint16_t>(port));
+      if (inet_pton(AF_INET, host.c_str(), &addr.sin_addr) != 1)
+        addr.sin_addr.s_addr = INADDR_ANY;
+      if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr),
+                 sizeof(addr)) < 0) {
+        ::close(listen_fd_);
+        throw std::runtime_error("bind() failed on port " +
+                                 std::to_string(port));
+      }
+      socklen_t len = sizeof(addr);
+      getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&addr), &len);
+      port_ = ntohs(addr.sin_port);
     }
-    socklen_t len = sizeof(addr);
-    getsockname(listen_fd_, reinterpret_cast<sockaddr*>(&addr), &len);
-    port_ = ntohs(addr.sin_port);
     if (::listen(listen_fd_, 128) < 0) {
       ::close(listen_fd_);
       throw std::runtime_error("listen() failed");

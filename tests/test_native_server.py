@@ -331,3 +331,25 @@ def test_debug_latency_and_heap_endpoints(native):
         snap = r2.json()
         assert snap["rss_kib"] > 0
         assert isinstance(snap["top"], list) and snap["top"]
+
+
+def test_native_server_ipv6(fake_client):
+    """Dual-stack: an IPv6 host literal binds AF_INET6 (IPv6-first
+    clusters); IPv4 behavior is unchanged (every other test)."""
+    fake_client.add_node(make_node("node-a"))
+    registry = SchedulerRegistry(fake_client)
+    registry.default._ensure_node("node-a")
+    app = make_app(registry)
+    fe = NativeFrontend(app, host="::1", port=0)
+    fe.start()
+    try:
+        with httpx.Client(base_url=f"http://[::1]:{fe.port}",
+                          timeout=10.0) as c:
+            assert c.get("/healthz").status_code == 200
+            pod = fake_client.create_pod(make_pod("p6", core=10))
+            r = c.post("/scheduler/filter",
+                       json={"pod": pod, "nodenames": ["node-a"]})
+            assert r.status_code == 200
+            assert r.json()["nodenames"] == ["node-a"]
+    finally:
+        fe.stop()
