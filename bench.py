@@ -534,11 +534,14 @@ def main():
     p.add_argument("--cards", type=int, default=8)
     p.add_argument("--policy", default="binpack",
                    choices=("binpack", "spread", "random"))
-    p.add_argument("--concurrency", type=int, default=4,
-                   help="in-flight pods in the load generator (4 measured "
-                        "best for BOTH throughput and latency: 2195 pods/s "
-                        "at p50 1.3 ms vs 16-way's ~1900 at ~4 ms — higher "
-                        "concurrency only adds queueing)")
+    p.add_argument("--concurrency", type=int, default=1,
+                   help="in-flight pods in the load generator. 1 measured "
+                        "best on the r2 build (MI355X box: 3078 pods/s at "
+                        "p50 0.24 ms vs 4-way's 2485 at 1.17 ms — the C++ "
+                        "fast paths got cheap enough that extra in-flight "
+                        "pods only buy GIL contention on the bind path); "
+                        "--kube-sim models the real single-scheduler "
+                        "regime with async binds")
     p.add_argument("--filter-threads", type=int, default=0)
     p.add_argument("--server", default="native",
                    choices=("native", "uvicorn"),
