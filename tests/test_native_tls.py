@@ -139,3 +139,34 @@ def test_throughput_parity_under_tls(pki, tls_stack):
                        json={"pod": pod, "nodenames": ["node-a"]})
             assert r.status_code == 200
         assert fe.stats()["filter_native"] - before == 50
+
+
+def test_ipv6_with_tls(pki, fake_client):
+    """Dual-stack + HTTPS together: TLS terminates on an AF_INET6 socket.
+    The PKI's SAN covers localhost, which resolves to ::1 here."""
+    fake_client.add_node(make_node("node-a"))
+    registry = SchedulerRegistry(fake_client)
+    registry.default._ensure_node("node-a")
+    app = make_app(registry)
+    fe = NativeFrontend(app, host="::1", port=0,
+                        tls_cert=pki["server_crt"], tls_key=pki["server_key"])
+    fe.start()
+    try:
+        # raw TLS over an IPv6 TCP connection (SNI/verification against
+        # the cert's DNS:localhost SAN)
+        import socket
+
+        ctx = ssl.create_default_context(cafile=pki["ca_crt"])
+        raw = socket.create_connection(("::1", fe.port))
+        with ctx.wrap_socket(raw, server_hostname="localhost") as tls:
+            tls.sendall(b"GET /healthz HTTP/1.1\r\nhost: a\r\n"
+                        b"connection: close\r\n\r\n")
+            data = b""
+            while True:
+                chunk = tls.recv(4096)
+                if not chunk:
+                    break
+                data += chunk
+        assert b"200 OK" in data and b'{"ok": true}' in data
+    finally:
+        fe.stop()

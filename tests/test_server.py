@@ -227,3 +227,16 @@ def test_pgpu_mode_end_to_end(fake_client):
     assert ext.bind(pod, "node-a").status_code == 200
     bound = fake_client.get_pod("default", "p")
     assert bound["metadata"]["annotations"]["elasticgpu.io/container-c0"] == "0,1"
+
+
+def test_hist_quantile_math():
+    """log2-us histogram quantile helper used by /debug/latency."""
+    from elastic_gpu_scheduler_amd.server.app import _hist_quantile
+
+    h = {"count": 0, "buckets": [(1, 0), (2, 0)]}
+    assert _hist_quantile(h, 0.5) is None
+    # 10 samples <=2us, 90 samples <=1024us
+    h = {"count": 100, "buckets": [(2, 10), (1024, 90)]}
+    assert _hist_quantile(h, 0.05) == 2
+    assert _hist_quantile(h, 0.5) == 1024
+    assert _hist_quantile(h, 0.99) == 1024
