@@ -209,7 +209,10 @@ def test_invalidation_churn_under_load():
                 try:
                     ok, _ = sch.assume(["n1"], pod)
                     if not ok:
-                        break
+                        # transient: the invalidator can evict between the
+                        # ensure and the fan-out — kube-scheduler requeues
+                        # unschedulable pods, so requeue here too
+                        continue
                     sch.bind("n1", client.get_pod(
                         "default", pod["metadata"]["name"]))
                     with bound_mu:
